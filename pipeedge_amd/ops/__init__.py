@@ -1,0 +1,107 @@
+"""Compute-op dispatch: hand-written CDNA4 HIP kernels on GPU, torch on CPU.
+
+Every model shard calls through this module. On a CUDA (ROCm) device the ops
+MUST run on the in-tree HIP extension (``pipeedge_amd/ops/hip``); if the
+extension is missing or an op is unimplemented, the call raises instead of
+silently falling back to eager PyTorch. Set ``PIPEEDGE_AMD_FORCE_TORCH=1`` to
+explicitly opt into the eager path on GPU (used for A/B comparisons only).
+
+The reference framework has no kernels at all - its
+hot path is eager HF modules; this module is where the MI355X build replaces
+them.
+"""
+import math
+import os
+from typing import Optional
+
+import torch
+from torch import Tensor
+
+from . import torch_ref
+
+_hip = None          # the loaded extension module, or None
+_hip_load_err = None  # exception from the load attempt, if any
+_hip_tried = False
+
+
+def _load_hip():
+    global _hip, _hip_load_err, _hip_tried
+    if _hip_tried:
+        return _hip
+    _hip_tried = True
+    try:
+        from . import hip_loader
+        _hip = hip_loader.load()
+    except Exception as exc:  # noqa: BLE001 - report it at use time
+        _hip_load_err = exc
+        _hip = None
+    return _hip
+
+
+def force_torch() -> bool:
+    return os.environ.get("PIPEEDGE_AMD_FORCE_TORCH", "0") == "1"
+
+
+def _use_hip(x: Tensor) -> bool:
+    """HIP-kernel path iff tensor is on GPU and the eager escape is not set."""
+    if not x.is_cuda or force_torch():
+        return False
+    if _load_hip() is None:
+        raise RuntimeError(
+            "pipeedge_amd HIP extension is required for GPU execution but "
+            f"failed to load: {_hip_load_err!r}. Build it with "
+            "`python setup.py build_ext --inplace` (or __graft_entry__.build()); "
+            "set PIPEEDGE_AMD_FORCE_TORCH=1 only to explicitly run eager torch."
+        )
+    return True
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def layernorm(x: Tensor, weight: Tensor, bias: Tensor, eps: float = 1e-6) -> Tensor:
+    if _use_hip(x):
+        return _hip.layernorm(x, weight, bias, eps)
+    return torch_ref.layernorm(x, weight, bias, eps)
+
+
+def add_layernorm(x: Tensor, residual: Tensor, weight: Tensor, bias: Tensor,
+                  eps: float = 1e-6) -> Tensor:
+    if _use_hip(x):
+        return _hip.add_layernorm(x, residual, weight, bias, eps)
+    return torch_ref.add_layernorm(x, residual, weight, bias, eps)
+
+
+_ACT_CODE = {None: 0, "gelu": 1, "tanh": 2}
+
+
+def linear(x: Tensor, weight_kn: Tensor, bias: Optional[Tensor] = None,
+           act: Optional[str] = None, residual: Optional[Tensor] = None) -> Tensor:
+    if _use_hip(x):
+        return _hip.linear(x, weight_kn, bias, _ACT_CODE[act], residual)
+    return torch_ref.linear(x, weight_kn, bias, act, residual)
+
+
+def attention_core(q: Tensor, k: Tensor, v: Tensor, scale: Optional[float] = None,
+                   attn_mask: Optional[Tensor] = None) -> Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _use_hip(q):
+        return _hip.attention(q, k, v, scale, attn_mask)
+    return torch_ref.attention_core(q, k, v, scale, attn_mask)
+
+
+def qkv_attention(qkv: Tensor, num_heads: int, scale: Optional[float] = None,
+                  attn_mask: Optional[Tensor] = None) -> Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(qkv.shape[-1] // 3 // num_heads)
+    if _use_hip(qkv):
+        return _hip.qkv_attention(qkv, num_heads, scale, attn_mask)
+    return torch_ref.qkv_attention(qkv, num_heads, scale, attn_mask)
+
+
+def patchify(pixels: Tensor, patch_size: int) -> Tensor:
+    if _use_hip(pixels):
+        return _hip.patchify(pixels, patch_size)
+    return torch_ref.patchify(pixels, patch_size)
