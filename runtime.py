@@ -1,0 +1,551 @@
+"""Pipeline-parallel inference runtime CLI (parity: reference runtime.py).
+
+Same CLI surface as the reference (positional rank/worldsize; -d/-s/--addr/
+--port/-c/-w/-m/-M/-b/-u; dataset group; user-scheduling -pt/-q/-r/-D;
+auto-scheduling -H/-sm/-sdt/-sd) with MI355X-native internals:
+
+- one process per GPU; activations stay GPU-resident and move over RCCL/xGMI
+  ("nccl" backend) - gloo only for the CPU control plane and CPU-only hosts;
+- the schedule broadcast carries everything needed for each rank to derive
+  its static wire schemas (no per-message headers);
+- quantization (QuantPipe) runs as stage send/recv transforms with a wire
+  payload sized for the configured max bitwidth, so adaptive bitwidth changes
+  never change the schema.
+"""
+import argparse
+import logging
+import os
+import queue
+import sys
+import threading
+import time
+from typing import List, Optional, Tuple, Union
+
+import numpy as np
+import torch
+
+import devices
+import model_cfg
+import monitoring
+from pipeedge_amd import models, quant
+from pipeedge_amd.comm import CMD_BATCH, CMD_SCHED, CMD_STOP
+from pipeedge_amd.comm.p2p import DistP2pContext
+from pipeedge_amd.models import spec as model_spec
+from pipeedge_amd.quant import (clamp_banner2019_gelu, clamp_banner2019_laplace,
+                                tensor_decode_outerdim, tensor_encode_outerdim)
+from pipeedge_amd.sched.scheduler import sched_pipeline
+from pipeedge_amd.utils.data import load_dataset_synthetic
+from pipeedge_amd.utils.threads import ThreadSafeCounter
+
+logger = logging.getLogger(__name__)
+
+# monitoring keys (parity: reference runtime.py:54-59)
+MONITORING_KEY_MODEL = "shard"
+MONITORING_KEY_OUTPUT = "output"
+MONITORING_KEY_QUANT_DECODE = "quant_decode"
+MONITORING_KEY_QUANT_ENCODE = "quant_encode"
+MONITORING_KEY_RECV = "recv"
+MONITORING_KEY_SEND = "send"
+
+ENV_WINDOW_SIZE = "WINDOW_SIZE"
+WINDOW_SIZE_DEFAULT = 10
+
+ENV_SEND_CONSTRAINT = "SEND_CONSTRAINT"
+ENV_ADAPTIVE_QUANT = "ADAPTIVE_QUANT"
+ADAPTIVE_QUANT_HEURISTIC = "HEURISTIC"
+ADAPTIVE_QUANT_HEURISTIC2 = "HEURISTIC2"
+ADAPTIVE_QUANT_CONTROLLER = "CONTROLLER"
+
+
+def get_window_size() -> int:
+    return int(os.getenv(ENV_WINDOW_SIZE, str(WINDOW_SIZE_DEFAULT)))
+
+
+# ---------------------------------------------------------------------------
+# command handling / schedule distribution
+# ---------------------------------------------------------------------------
+
+sched_q = queue.Queue()
+batch_q = queue.Queue()
+stop_event = threading.Event()
+
+
+def handle_cmd(cmd: int, tensors: Tuple[torch.Tensor, ...]) -> None:
+    """Process control-plane commands (parity: reference runtime.py:406-415)."""
+    if cmd == CMD_STOP:
+        logger.info("handle_cmd: stop")
+        stop_event.set()
+    elif cmd == CMD_SCHED:
+        logger.info("handle_cmd: sched")
+        sched_q.put((tensors[0].tolist(),      # stage layers
+                     tensors[1].tolist(),      # stage quant
+                     tensors[2].tolist(),      # stage ranks
+                     int(tensors[3].item()),   # data rank
+                     int(tensors[4].item())))  # seq len (0 = model default)
+    elif cmd == CMD_BATCH:
+        batch_q.put(int(tensors[0].item()))
+    else:
+        logger.warning("handle_cmd: unknown command: %d", cmd)
+
+
+def parse_yaml_sched(sched: List[dict], hosts: Optional[List[str]]) \
+        -> Tuple[List[Tuple[int, int]], List[int]]:
+    """Turn scheduler YAML output into (stage_layers, stage_ranks)
+    (parity: reference runtime.py:260-289)."""
+    stage_layers = []
+    stage_ranks = []
+    if len(sched) == 0:
+        raise RuntimeError("No viable schedule found")
+    for stage in sched:
+        assert len(stage) == 1
+        host = list(stage.keys())[0]
+        layers = stage[host]
+        stage_layers.append((int(layers[0]), int(layers[1])))
+        if hosts:
+            try:
+                stage_ranks.append(hosts.index(host))
+            except ValueError:
+                logger.error("Scheduling: host not found in hosts list: %s", host)
+                raise
+        else:
+            try:
+                stage_ranks.append(int(host))
+            except ValueError:
+                logger.error("Scheduling: 'hosts' not specified and host is "
+                             "not an int: %s", host)
+                raise
+    return stage_layers, stage_ranks
+
+
+def get_pipeline_sched(world_size: int, hosts: Optional[List[str]],
+                       partition: Optional[List[Tuple[int, int]]],
+                       quant: Optional[List[int]],
+                       rank_order: Optional[List[int]], model_name: str,
+                       ubatch_size: int, dtype_str: str,
+                       s_models_file: Optional[str],
+                       s_dev_types_file: Optional[str],
+                       s_dev_file: Optional[str]) \
+        -> Tuple[List[Tuple[int, int]], List[int], List[int]]:
+    """Resolve the pipeline schedule (parity: reference runtime.py:291-356)."""
+    if partition:
+        stage_layers = partition
+        stage_quant = quant if quant else [0] * len(stage_layers)
+        stage_ranks = rank_order if rank_order else list(range(len(stage_layers)))
+    elif world_size <= 1:
+        # degenerate single-stage pipeline (reference runtime.py:324-329)
+        stage_layers = [(1, model_cfg.get_model_layers(model_name))]
+        stage_quant = quant if quant else [0]
+        stage_ranks = rank_order if rank_order else [0]
+    elif s_models_file and s_dev_types_file and s_dev_file:
+        sched = sched_pipeline(model_name, 2, 2, ubatch_size, dtype=dtype_str,
+                               models_file=s_models_file,
+                               dev_types_file=s_dev_types_file,
+                               dev_file=s_dev_file)
+        stage_layers, stage_ranks = parse_yaml_sched(sched, hosts)
+        stage_quant = quant if quant else [0] * len(stage_layers)
+    else:
+        raise RuntimeError("Scheduling: must specify either -pt or all of "
+                           "-sm/-sdt/-sd (or run with world size <= 1)")
+    assert len(stage_layers) == len(stage_quant) == len(stage_ranks)
+    logger.info("Scheduling: stage-to-layer mapping: %s", stage_layers)
+    logger.info("Scheduling: stage output quantization: %s", stage_quant)
+    logger.info("Scheduling: stage-to-rank mapping: %s", stage_ranks)
+    return stage_layers, stage_quant, stage_ranks
+
+
+# ---------------------------------------------------------------------------
+# quantization stage transforms (replace the reference's forward hooks,
+# runtime.py:73-119; wire format unchanged)
+# ---------------------------------------------------------------------------
+
+def make_quant_send_transform(module, pad_words_per_tensor: List[int]):
+    """Encode stage outputs with the module's current (adaptive) quant_bit."""
+    def transform(outputs: models.ModuleShardData) -> List[torch.Tensor]:
+        monitoring.iteration_start(MONITORING_KEY_QUANT_ENCODE)
+        if isinstance(outputs, torch.Tensor):
+            outputs = (outputs,)
+        quant_bit = int(module.quant_bit.item())
+        wire: List[torch.Tensor] = []
+        for tensor, pad_words in zip(outputs, pad_words_per_tensor):
+            if quant_bit > 0:
+                clamp = clamp_banner2019_laplace if tensor.min() < 0.2 \
+                    else clamp_banner2019_gelu
+                tensor = clamp(tensor, quant_bit)
+            wire += tensor_encode_outerdim(tensor, quant_bit, pad_words=pad_words)
+        n_items = models.get_microbatch_size(outputs[0]) if quant_bit > 0 else 0
+        monitoring.iteration(MONITORING_KEY_QUANT_ENCODE, work=n_items,
+                             accuracy=quant_bit)
+        return wire
+    return transform
+
+
+def make_quant_recv_transform(dtype: torch.dtype):
+    """Decode 5-tensor groups back into stage inputs."""
+    def transform(wire: List[torch.Tensor]) -> models.ModuleShardData:
+        monitoring.iteration_start(MONITORING_KEY_QUANT_DECODE)
+        assert len(wire) % 5 == 0 and len(wire) >= 5
+        quant_bit = int(wire[4].view(-1)[0].item())
+        tensors = []
+        for i in range(len(wire) // 5):
+            group = wire[i * 5:i * 5 + 5]
+            tensors.append(tensor_decode_outerdim(group).to(dtype))
+        n_items = models.get_microbatch_size(tensors[0]) if quant_bit > 0 else 0
+        monitoring.iteration(MONITORING_KEY_QUANT_DECODE, work=n_items,
+                             accuracy=quant_bit)
+        outputs = tensors[0] if len(tensors) == 1 else tuple(tensors)
+        return outputs
+    return transform
+
+
+def cast_recv_transform(dtype: torch.dtype):
+    """Un-quantized inbound edge: cast to the compute dtype."""
+    def transform(wire: List[torch.Tensor]) -> models.ModuleShardData:
+        tensors = [t.to(dtype) if t.is_floating_point() else t for t in wire]
+        return tensors[0] if len(tensors) == 1 else tuple(tensors)
+    return transform
+
+
+# p2p comm monitoring hooks (parity: reference runtime.py:219-234)
+def p2p_pre_hook_monitor(key: str) -> None:
+    monitoring.iteration_start(key)
+
+
+def p2p_post_hook_monitor(key: str, tensors: List[torch.Tensor]) -> None:
+    n_bytes = sum(t.numel() * t.element_size() for t in tensors)
+    mbits = n_bytes * 8 / 1e6
+    monitoring.iteration(key, work=mbits)
+
+
+# ---------------------------------------------------------------------------
+# results handling (parity: reference runtime.py:236-257)
+# ---------------------------------------------------------------------------
+
+results_counter = ThreadSafeCounter()
+label_queue = queue.Queue()
+
+
+def handle_results(tensors: models.ModuleShardData) -> None:
+    """Process pipeline outputs: accuracy bookkeeping + counting."""
+    if isinstance(tensors, tuple):
+        tensors = tensors[0]
+    n_items = models.get_microbatch_size(tensors)
+    monitoring.iteration_start(MONITORING_KEY_OUTPUT)
+    acc = 0
+    if not label_queue.empty() and tensors.dim() == 2:
+        labels = label_queue.get()
+        pred = tensors.float().argmax(dim=-1).cpu()
+        acc = int((pred == labels[:pred.shape[0]]).sum().item())
+    monitoring.iteration(MONITORING_KEY_OUTPUT, work=n_items, accuracy=acc)
+    results_counter.add(n_items)
+
+
+# ---------------------------------------------------------------------------
+# pipeline drivers
+# ---------------------------------------------------------------------------
+
+def _compute_dtype(device: torch.device) -> torch.dtype:
+    return torch.bfloat16 if device.type == "cuda" else torch.float32
+
+
+def load_dataset(dataset_cfg: dict, model_name: str, batch_size: int,
+                 ubatch_size: int):
+    """Dataset dispatch (reference runtime.py:358-401). Default: synthetic
+    random inputs (offline environment)."""
+    config = model_cfg.get_model_config(model_name)
+    name = dataset_cfg.get("name")
+    if name == "CoLA":
+        from pipeedge_amd.utils.data import load_dataset_glue
+        return load_dataset_glue(model_name, "cola", dataset_cfg.get("split", "train"),
+                                 batch_size, indices=dataset_cfg.get("indices"))
+    if name == "ImageNet":
+        from pipeedge_amd.utils.data import load_dataset_imagenet
+        return load_dataset_imagenet(dataset_cfg["root"],
+                                     dataset_cfg.get("split", "train"),
+                                     batch_size,
+                                     indices=dataset_cfg.get("indices"))
+    return load_dataset_synthetic(config, batch_size)
+
+
+def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
+                     model_file: Optional[str], batch_size: int,
+                     ubatch_size: int, partition, quant_list, rank_order,
+                     data_rank: int, hosts, dataset_cfg: dict,
+                     sched_models_file, sched_dev_types_file, sched_dev_file,
+                     num_batches: int = 1) -> None:
+    """Run the P2P pipeline (parity: reference runtime.py:418-511)."""
+    monitoring.init(MONITORING_KEY_MODEL, get_window_size(),
+                    work_type="tensors", acc_type="layers")
+    monitoring.add_key(MONITORING_KEY_OUTPUT, work_type="classifications",
+                       acc_type="correct")
+    monitoring.add_key(MONITORING_KEY_QUANT_DECODE, work_type="tensors",
+                       acc_type="bits")
+    monitoring.add_key(MONITORING_KEY_QUANT_ENCODE, work_type="tensors",
+                       acc_type="bits")
+    monitoring.add_key(MONITORING_KEY_RECV, work_type="Mbits")
+    monitoring.add_key(MONITORING_KEY_SEND, work_type="Mbits")
+
+    device = devices.DEVICE if devices.DEVICE is not None else torch.device("cpu")
+    dtype = _compute_dtype(device)
+    dtype_str = f"torch.{str(dtype).split('.')[-1]}"
+
+    with DistP2pContext(world_size, rank, handle_cmd) as dist_ctx:
+        # distribute the schedule (reference runtime.py:431-448)
+        if rank == 0:
+            stage_layers, stage_quant, stage_ranks = get_pipeline_sched(
+                world_size, hosts, partition, quant_list, rank_order,
+                model_name, ubatch_size, dtype_str, sched_models_file,
+                sched_dev_types_file, sched_dev_file)
+            # seq_len: data-dependent for text models; broadcast with schedule
+            seq_len = 0
+            if model_cfg.get_model_config(model_name).model_type == "bert":
+                dataset_probe = load_dataset(dataset_cfg, model_name, 1, 1)
+                seq_len = dataset_probe[0][0].shape[-1]
+            logger.info("Scheduling: data rank: %s", data_rank)
+            dist_ctx.cmd_broadcast(CMD_SCHED, (
+                torch.tensor(stage_layers),
+                torch.tensor(stage_quant),
+                torch.tensor(stage_ranks),
+                torch.tensor(data_rank),
+                torch.tensor(seq_len)))
+        else:
+            logger.info("Waiting for schedule")
+            stage_layers, stage_quant, stage_ranks, data_rank, seq_len = \
+                sched_q.get()
+            stage_layers = [tuple(sl) for sl in stage_layers]
+
+        try:
+            stage = stage_ranks.index(rank)
+        except ValueError:
+            stage = None
+
+        module = None
+        if stage is not None:
+            module = model_cfg.module_shard_factory(
+                model_name, model_file, stage_layers[stage][0],
+                stage_layers[stage][1], stage, dtype=dtype)
+            module.register_buffer("quant_bit",
+                                   torch.tensor(stage_quant[stage]),
+                                   persistent=False)
+            module.register_forward_pre_hook(
+                lambda _m, _i: monitoring.iteration_start(MONITORING_KEY_MODEL))
+
+            def _fwd_monitor(mod, _inputs, outputs):
+                n_items = models.get_microbatch_size(outputs, verify=True)
+                n_layers = mod.shard_config.layer_end - \
+                    mod.shard_config.layer_start + 1
+                monitoring.iteration(MONITORING_KEY_MODEL, work=n_items,
+                                     accuracy=n_layers)
+            module.register_forward_hook(_fwd_monitor)
+
+        stage_ctx = model_cfg.dist_p2p_pipeline_stage_factory(
+            stage_ranks, data_rank, rank, stage, module, handle_results,
+            model_name=model_name, ubatch_size=ubatch_size,
+            stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype)
+
+        # attach quant / cast transforms to the stage's edges
+        config = model_cfg.get_model_config(model_name)
+        total_layers = model_cfg.get_model_layers(model_name)
+        if stage is not None:
+            # outbound: quantize unless we're the last stage
+            if stage_layers[stage][1] < total_layers and stage_quant[stage] > 0:
+                out_specs = model_spec.stage_boundary_spec(
+                    config, stage_layers[stage][1], total_layers, ubatch_size,
+                    dtype, seq_len or None)
+                pad_words = [_pad_words(s, stage_quant[stage])
+                             for s in out_specs]
+                stage_ctx.send_transform = make_quant_send_transform(
+                    module, pad_words)
+            # inbound: dequantize unless we're stage 0
+            if stage > 0 and stage_quant[stage - 1] > 0:
+                stage_ctx.recv_transform = make_quant_recv_transform(dtype)
+            else:
+                stage_ctx.recv_transform = cast_recv_transform(dtype)
+        stage_ctx.register_recv_pre_hook(p2p_pre_hook_monitor,
+                                         (MONITORING_KEY_RECV,))
+        stage_ctx.register_recv_post_hook(p2p_post_hook_monitor,
+                                          (MONITORING_KEY_RECV,))
+        stage_ctx.register_send_pre_hook(p2p_pre_hook_monitor,
+                                         (MONITORING_KEY_SEND,))
+        stage_ctx.register_send_post_hook(p2p_post_hook_monitor,
+                                          (MONITORING_KEY_SEND,))
+
+        if rank == data_rank:
+            dataset = load_dataset(dataset_cfg, model_name, batch_size,
+                                   ubatch_size)
+            ubatches = []
+            labels = []
+            for start in range(0, len(dataset), ubatch_size):
+                items = [dataset[i] for i in
+                         range(start, min(start + ubatch_size, len(dataset)))]
+                data = torch.stack([it[0] for it in items])
+                lbl = torch.stack([torch.as_tensor(it[1]) for it in items])
+                if data.is_floating_point():
+                    data = data.to(dtype)
+                ubatches.append(data)
+                labels.append(lbl)
+            tik_data = time.time()
+            start_count = results_counter.value
+            for _ in range(num_batches):
+                for lbl in labels:
+                    label_queue.put(lbl)
+                dist_ctx.cmd_broadcast(CMD_BATCH,
+                                       (torch.tensor(len(ubatches)),))
+                stage_ctx.data_run_batch(ubatches)
+            results_counter.wait_gte(start_count + len(dataset) * num_batches)
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+            tok_data = time.time()
+            latency = tok_data - tik_data
+            throughput = batch_size * num_batches / latency
+            logger.info("Latency is %f, throughput is %f items/sec",
+                        latency, throughput)
+            dist_ctx.cmd_broadcast(CMD_STOP)
+            stop_event.set()
+        elif stage is not None:
+            while not stop_event.is_set():
+                try:
+                    n_ubatch = batch_q.get(timeout=0.1)
+                except queue.Empty:
+                    continue
+                stage_ctx.stage_run_batch(n_ubatch)
+        else:
+            stop_event.wait()
+    monitoring.finish()
+
+
+def _pad_words(tensor_spec, quant_bit: int) -> Optional[int]:
+    if quant_bit <= 0:
+        return None
+    import math
+    per_elem = 1
+    for d in tensor_spec.shape[1:]:
+        per_elem *= d
+    return math.ceil(per_elem / (32 // quant_bit))
+
+
+def run_pipeline_rpc(*args, **kwargs) -> None:
+    """RPC transport (reference runtime.py:514-578). See comm/rpc.py."""
+    from pipeedge_amd.comm import rpc as rpc_mod
+    rpc_mod.run_pipeline_rpc(*args, **kwargs)
+
+
+def init_env(device: Optional[str], net_addr: str, net_port: int,
+             net_ifname: str) -> None:
+    """Initialize the runtime environment (reference runtime.py:581-602)."""
+    devices.initialize(device)
+    os.environ.setdefault("MASTER_ADDR", net_addr)
+    os.environ.setdefault("MASTER_PORT", str(net_port))
+    if net_ifname and net_ifname != "lo0":
+        os.environ.setdefault("GLOO_SOCKET_IFNAME", net_ifname)
+        os.environ.setdefault("TP_SOCKET_IFNAME", net_ifname)
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser(
+        description="MI355X Pipeline Parallelism Runtime",
+        formatter_class=argparse.ArgumentDefaultsHelpFormatter)
+    parser.add_argument("rank", type=int, help="the rank for the current node")
+    parser.add_argument("worldsize", type=int, help="the world size")
+    parser.add_argument("-d", "--device", type=str, default=None,
+                        help="compute device: 'cpu', 'cuda', 'cuda:1', ...")
+    parser.add_argument("-s", "--socket-ifname", type=str, default="lo0",
+                        help="socket interface name")
+    parser.add_argument("--addr", type=str, default="127.0.0.1",
+                        help="ip address for the master node")
+    parser.add_argument("--port", type=int, default=29500,
+                        help="communication port for the master node")
+    parser.add_argument("-c", "--comm", type=str, default="p2p",
+                        choices=["p2p", "rpc"],
+                        help="the communication implementation")
+    parser.add_argument("-w", "--worker-threads", default=16, type=int,
+                        help="worker threads for the 'rpc' backend")
+    parser.add_argument("-m", "--model-name", type=str,
+                        default="google/vit-base-patch16-224",
+                        choices=model_cfg.get_model_names(),
+                        help="the neural network model for loading")
+    parser.add_argument("-M", "--model-file", type=str,
+                        help="the model weights file")
+    parser.add_argument("-b", "--batch-size", default=64, type=int)
+    parser.add_argument("-u", "--ubatch-size", default=8, type=int)
+    parser.add_argument("--num-batches", default=1, type=int,
+                        help="number of batches to stream (throughput runs)")
+    dset = parser.add_argument_group("Dataset arguments")
+    dset.add_argument("--dataset-name", type=str,
+                      choices=["CoLA", "ImageNet", "synthetic"],
+                      help="dataset to use (default: synthetic random)")
+    dset.add_argument("--dataset-root", type=str,
+                      help="dataset root directory")
+    dset.add_argument("--dataset-split", default="train", type=str)
+    dset.add_argument("--dataset-indices-file", default=None, type=str)
+    dset.add_argument("--dataset-shuffle", type=bool, nargs="?", const=True,
+                      default=False)
+    usched = parser.add_argument_group("User-defined scheduling")
+    usched.add_argument("-pt", "--partition", type=str,
+                        help="comma-delimited layer pairs, e.g. '1,24,25,48'")
+    usched.add_argument("-q", "--quant", type=str,
+                        help="comma-delimited quantization bits per stage")
+    usched.add_argument("-r", "--rank-order", type=str, default=None,
+                        help="comma-delimited ranks in stage order")
+    usched.add_argument("-D", "--data-rank", type=int, default=0,
+                        help="rank where inputs are loaded / outputs handled")
+    asched = parser.add_argument_group("Automated scheduling")
+    asched.add_argument("-H", "--hosts", type=str,
+                        help="comma-delimited hosts in rank order")
+    asched.add_argument("-sm", "--sched-models-file", default=None, type=str)
+    asched.add_argument("-sdt", "--sched-dev-types-file", default=None, type=str)
+    asched.add_argument("-sd", "--sched-dev-file", default=None, type=str)
+    args = parser.parse_args()
+
+    if args.dataset_indices_file is None:
+        indices = None
+    elif args.dataset_indices_file.endswith(".pt"):
+        indices = torch.load(args.dataset_indices_file)
+    else:
+        indices = np.load(args.dataset_indices_file)
+    dataset_cfg = {
+        "name": args.dataset_name,
+        "root": args.dataset_root,
+        "split": args.dataset_split,
+        "indices": indices,
+        "shuffle": args.dataset_shuffle,
+    }
+
+    partition = None
+    if args.partition is not None:
+        parts = [int(i) for i in args.partition.split(",")]
+        assert len(parts) % 2 == 0
+        partition = [(parts[i], parts[i + 1]) for i in range(0, len(parts), 2)]
+    quant_list = None if args.quant is None else \
+        [int(i) for i in args.quant.split(",")]
+    rank_order = None if args.rank_order is None else \
+        [int(i) for i in args.rank_order.split(",")]
+    hosts = None if args.hosts is None else args.hosts.split(",")
+
+    tik = time.time()
+    init_env(args.device, args.addr, args.port, args.socket_ifname)
+    logger.info("Device: %s", devices.DEVICE)
+    if args.comm == "p2p":
+        run_pipeline_p2p(args.worldsize, args.rank, args.model_name,
+                         args.model_file, args.batch_size, args.ubatch_size,
+                         partition, quant_list, rank_order, args.data_rank,
+                         hosts, dataset_cfg, args.sched_models_file,
+                         args.sched_dev_types_file, args.sched_dev_file,
+                         num_batches=args.num_batches)
+    else:
+        run_pipeline_rpc(args.worldsize, args.rank, args.model_name,
+                         args.model_file, args.batch_size, args.ubatch_size,
+                         partition, quant_list, rank_order, args.data_rank,
+                         hosts, dataset_cfg, args.sched_models_file,
+                         args.sched_dev_types_file, args.sched_dev_file,
+                         args.worker_threads)
+    tok = time.time()
+    logger.info("Total program execution time = %f", tok - tik)
+
+
+if __name__ == "__main__":
+    logging.basicConfig(filename="runtime.log", level=logging.DEBUG)
+    console_hndlr = logging.StreamHandler(sys.stdout)
+    console_hndlr.setFormatter(logging.Formatter(fmt="%(message)s"))
+    console_hndlr.setLevel(logging.INFO)
+    logging.getLogger().addHandler(console_hndlr)
+    main()
