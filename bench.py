@@ -1,0 +1,192 @@
+"""Flagship benchmark: ViT-Base/224 pipelined inference throughput.
+
+Measures the BASELINE.json metric - images/sec (whole node) for
+google/vit-base-patch16-224 at microbatch=8, batch=64, N-stage pipeline over
+N GPUs (RCCL p2p over xGMI; one rank per GPU) - on synthetic ImageNet-shaped
+inputs and random-init weights (no network in this environment).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One step = one batch of ``--batch-size`` images streamed through the pipeline
+as microbatches. W untimed warmup steps, then exactly K timed steps bracketed
+by barrier + torch.cuda.synchronize on both sides; elapsed = MAX over ranks;
+rank 0 prints ONE JSON line.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import devices  # noqa: E402
+import model_cfg  # noqa: E402
+from pipeedge_amd.utils.data import load_dataset_synthetic  # noqa: E402
+
+BASELINE_IMG_PER_SEC = 0.61  # ViT-Base/224 fp32 u=8, 1 edge device
+
+
+def even_partition(total_layers: int, n_stages: int):
+    """Split [1, total_layers] into n_stages contiguous near-even ranges."""
+    base = total_layers // n_stages
+    rem = total_layers % n_stages
+    parts = []
+    start = 1
+    for s in range(n_stages):
+        size = base + (1 if s < rem else 0)
+        parts.append((start, start + size - 1))
+        start += size
+    return parts
+
+
+def run_rank(args, rank: int, world_size: int) -> dict:
+    import torch.distributed as dist
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local_rank)
+        devices.initialize(f"cuda:{local_rank}")
+    else:
+        devices.initialize("cpu")
+    device = devices.DEVICE
+    dtype = torch.bfloat16 if use_cuda else torch.float32
+
+    distributed = world_size > 1
+    if distributed:
+        backend = "nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
+
+    model_name = args.model
+    total_layers = model_cfg.get_model_layers(model_name)
+    stage_layers = even_partition(total_layers, world_size)
+    stage_quant = [0] * world_size
+    stage_ranks = list(range(world_size))
+    stage = rank
+    data_rank = 0
+
+    module = model_cfg.module_shard_factory(
+        model_name, args.model_file, stage_layers[stage][0],
+        stage_layers[stage][1], stage, dtype=dtype)
+
+    results = []
+    stage_ctx = model_cfg.dist_p2p_pipeline_stage_factory(
+        stage_ranks, data_rank, rank, stage, module, results.append,
+        model_name=model_name, ubatch_size=args.ubatch_size,
+        stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype)
+
+    # synthetic data, resident on-device before the timed region
+    config = model_cfg.get_model_config(model_name)
+    dataset = load_dataset_synthetic(config, args.batch_size)
+    n_ubatch = args.batch_size // args.ubatch_size
+    ubatches = []
+    if rank == data_rank:
+        for u in range(n_ubatch):
+            items = [dataset[i][0] for i in range(u * args.ubatch_size,
+                                                  (u + 1) * args.ubatch_size)]
+            data = torch.stack(items)
+            if data.is_floating_point():
+                data = data.to(dtype)
+            ubatches.append(data.to(device))
+
+    def one_step():
+        if rank == data_rank:
+            results.clear()
+            stage_ctx.data_run_batch(ubatches)
+            assert len(results) == n_ubatch
+        else:
+            stage_ctx.stage_run_batch(n_ubatch)
+
+    def barrier_sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier_sync()
+    tik = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    barrier_sync()
+    tok = time.perf_counter()
+    elapsed = tok - tik
+
+    # elapsed = MAX over ranks
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    out = None
+    if rank == 0:
+        images = args.batch_size * args.steps
+        value = images / elapsed
+        out = {
+            "metric": "images/sec (whole node) ViT-Base/224 microbatch=8 "
+                      "at 1/2/4/8-stage pipeline",
+            "value": round(value, 3),
+            "unit": "images/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": round(value / BASELINE_IMG_PER_SEC, 2),
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": args.batch_size,
+                "ubatch_size": args.ubatch_size,
+                "seq_len": 197,
+                "parallelism": f"pp{world_size}",
+                "partition": [list(p) for p in stage_layers],
+                "weights": "random-init",
+            },
+        }
+    if distributed:
+        dist.barrier()
+        dist.destroy_process_group()
+    return out
+
+
+def main():
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument("--gpus", type=int, default=1,
+                        help="pipeline stages (= world size when distributed)")
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--batch-size", type=int, default=64)
+    parser.add_argument("--ubatch-size", type=int, default=8)
+    parser.add_argument("--model", type=str,
+                        default="google/vit-base-patch16-224")
+    parser.add_argument("--model-file", type=str, default=None)
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    if world_size == 1 and args.gpus > 1:
+        raise SystemExit("for --gpus N > 1, launch via torch.distributed.run "
+                         "with --nproc-per-node N")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+
+    out = run_rank(args, rank, world_size)
+    if out is not None:
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
