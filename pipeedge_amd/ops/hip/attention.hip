@@ -1,0 +1,221 @@
+// Fused multi-head attention over a packed QKV projection (bf16).
+//
+// ctx[b,s,h*dh+d] = softmax_s'(Q.K^T * scale) @ V for qkv[B,S,3D] with
+// columns [q | k | v], D = H*dh. Replaces the reference's HF ViTSelfAttention
+// (reference vit.py:45,60) - no permute/reshape
+// kernels, no S x S score materialization in HBM.
+//
+// Design (flash-style, CDNA4): one workgroup per (batch*head, 64-row Q
+// chunk); 4 waves x 16 Q rows each; KV tiles of 64 positions staged in LDS
+// (K natural [s'][d] for the B-fragment's contiguous-k reads, V transposed
+// [d][s']); online softmax with per-row running max/denominator in
+// registers; P goes through a wave-private LDS tile to re-shape from the
+// MFMA C layout to the A layout. Head dims up to DHP are supported by
+// zero-padding LDS columns (dh=64 -> DHP=64; dh=80 ViT-Huge -> DHP=96).
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int KVT = 64;   // kv positions per tile
+constexpr int QCH = 64;   // q rows per workgroup
+constexpr int PAD = 8;
+
+template <int DHP>
+__global__ __launch_bounds__(256) void qkv_attention_kernel(
+        const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
+        int B, int S, int H, int dh, float scale) {
+    __shared__ __bf16 Qs[QCH][DHP + PAD];
+    __shared__ __bf16 Ks[KVT][DHP + PAD];
+    __shared__ __bf16 Vs[DHP][KVT + PAD];      // transposed
+    __shared__ __bf16 Ps[4][16][KVT + PAD];    // per-wave P tile
+
+    const int bh = blockIdx.x;
+    const int b = bh / H;
+    const int h = bh % H;
+    const int q0 = blockIdx.y * QCH;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int D = H * dh;
+    const long row_stride = 3L * D;
+    const __bf16* q_base = qkv + (long)b * S * row_stride + h * dh;
+    const __bf16* k_base = q_base + D;
+    const __bf16* v_base = q_base + 2L * D;
+
+    // ---- stage Q chunk (once) ----
+    for (int piece = tid; piece < QCH * DHP / 8; piece += 256) {
+        const int r = piece / (DHP / 8);
+        const int c = (piece % (DHP / 8)) * 8;
+        bf16x8 v = {};
+        const int gs = q0 + r;
+        if (gs < S && c < dh) {
+            if (c + 7 < dh) {
+                v = *reinterpret_cast<const bf16x8*>(
+                    q_base + (long)gs * row_stride + c);
+            } else {
+                for (int j = 0; j < 8 && c + j < dh; ++j)
+                    v[j] = q_base[(long)gs * row_stride + c + j];
+            }
+        }
+        *reinterpret_cast<bf16x8*>(&Qs[r][c]) = v;
+    }
+
+    f32x4 o_acc[DHP / 16] = {};
+    float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+    float l_run[4] = {};
+
+    const int n_tiles = (S + KVT - 1) / KVT;
+    for (int t = 0; t < n_tiles; ++t) {
+        const int s0 = t * KVT;
+        __syncthreads();
+        // ---- stage K tile (natural) + V tile (transposed) ----
+        for (int piece = tid; piece < KVT * DHP / 8; piece += 256) {
+            const int r = piece / (DHP / 8);
+            const int c = (piece % (DHP / 8)) * 8;
+            const int gs = s0 + r;
+            bf16x8 kv_ = {};
+            bf16x8 vv = {};
+            if (gs < S && c < dh) {
+                if (c + 7 < dh) {
+                    kv_ = *reinterpret_cast<const bf16x8*>(
+                        k_base + (long)gs * row_stride + c);
+                    vv = *reinterpret_cast<const bf16x8*>(
+                        v_base + (long)gs * row_stride + c);
+                } else {
+                    for (int j = 0; j < 8 && c + j < dh; ++j) {
+                        kv_[j] = k_base[(long)gs * row_stride + c + j];
+                        vv[j] = v_base[(long)gs * row_stride + c + j];
+                    }
+                }
+            }
+            *reinterpret_cast<bf16x8*>(&Ks[r][c]) = kv_;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) Vs[c + j][r] = vv[j];
+        }
+        __syncthreads();
+
+        // ---- scores: S16x64 = Q16xDHP . K^T ----
+        f32x4 sfrag[KVT / 16] = {};
+        #pragma unroll
+        for (int kk = 0; kk < DHP; kk += 32) {
+            const int kbase = kk + (lane >> 4) * 8;
+            const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &Qs[wid * 16 + (lane & 15)][kbase]);
+            #pragma unroll
+            for (int nf = 0; nf < KVT / 16; ++nf) {
+                const bf16x8 bb = *reinterpret_cast<const bf16x8*>(
+                    &Ks[nf * 16 + (lane & 15)][kbase]);
+                sfrag[nf] = mfma_16x16x32_bf16(a, bb, sfrag[nf]);
+            }
+        }
+        // scale + mask invalid kv columns
+        #pragma unroll
+        for (int nf = 0; nf < KVT / 16; ++nf) {
+            const bool valid = s0 + nf * 16 + (lane & 15) < S;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                sfrag[nf][r] = valid ? sfrag[nf][r] * scale : -1e30f;
+        }
+
+        // ---- online softmax update (per C-fragment row r) ----
+        float p_local[KVT / 16][4];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float mt = -1e30f;
+            #pragma unroll
+            for (int nf = 0; nf < KVT / 16; ++nf)
+                mt = fmaxf(mt, sfrag[nf][r]);
+            mt = group16_reduce(mt, MaxOp());
+            const float m_new = fmaxf(m_run[r], mt);
+            const float alpha = __expf(m_run[r] - m_new);
+            float rowsum = 0.0f;
+            #pragma unroll
+            for (int nf = 0; nf < KVT / 16; ++nf) {
+                const float p = __expf(sfrag[nf][r] - m_new);
+                p_local[nf][r] = p;
+                rowsum += p;
+            }
+            rowsum = group16_reduce(rowsum, SumOp());
+            l_run[r] = l_run[r] * alpha + rowsum;
+            m_run[r] = m_new;
+            #pragma unroll
+            for (int nfo = 0; nfo < DHP / 16; ++nfo)
+                o_acc[nfo][r] *= alpha;
+        }
+        // P -> wave-private LDS (C layout -> A layout)
+        #pragma unroll
+        for (int nf = 0; nf < KVT / 16; ++nf)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                Ps[wid][(lane >> 4) * 4 + r][nf * 16 + (lane & 15)] =
+                    f2bf(p_local[nf][r]);
+
+        // ---- PV accumulate: O16xDHP += P16x64 . V64xDHP ----
+        #pragma unroll
+        for (int kk = 0; kk < KVT; kk += 32) {
+            const int kbase = kk + (lane >> 4) * 8;
+            const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &Ps[wid][lane & 15][kbase]);
+            #pragma unroll
+            for (int nfo = 0; nfo < DHP / 16; ++nfo) {
+                const bf16x8 bb = *reinterpret_cast<const bf16x8*>(
+                    &Vs[nfo * 16 + (lane & 15)][kbase]);
+                o_acc[nfo] = mfma_16x16x32_bf16(a, bb, o_acc[nfo]);
+            }
+        }
+    }
+
+    // ---- epilogue: normalize and store ctx rows ----
+    #pragma unroll
+    for (int nfo = 0; nfo < DHP / 16; ++nfo) {
+        const int d = nfo * 16 + (lane & 15);
+        if (d >= dh) continue;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int srow = q0 + wid * 16 + (lane >> 4) * 4 + r;
+            if (srow >= S) continue;
+            const float denom = l_run[r] > 0.0f ? l_run[r] : 1.0f;
+            ctx[((long)b * S + srow) * D + h * dh + d] =
+                f2bf(o_acc[nfo][r] / denom);
+        }
+    }
+}
+
+}  // namespace
+
+torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
+                                 double scale) {
+    TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16,
+                "attention: bf16 GPU tensor required");
+    TORCH_CHECK(qkv.dim() == 3, "attention: qkv must be [B, S, 3D]");
+    auto x = qkv.contiguous();
+    const int B = x.size(0), S = x.size(1);
+    const int D = x.size(2) / 3;
+    TORCH_CHECK(x.size(2) == 3 * D);
+    const int H = (int)num_heads;
+    TORCH_CHECK(D % H == 0);
+    const int dh = D / H;
+    auto ctx = torch::empty({B, S, D}, x.options());
+    dim3 grid(B * H, (S + QCH - 1) / QCH);
+    const auto* in = reinterpret_cast<const __bf16*>(x.data_ptr());
+    auto* out = reinterpret_cast<__bf16*>(ctx.data_ptr());
+    if (dh <= 64) {
+        hipLaunchKernelGGL((qkv_attention_kernel<64>), grid, dim3(256), 0,
+                           pe_current_stream(), in, out, B, S, H, dh,
+                           (float)scale);
+    } else if (dh <= 96) {
+        hipLaunchKernelGGL((qkv_attention_kernel<96>), grid, dim3(256), 0,
+                           pe_current_stream(), in, out, B, S, H, dh,
+                           (float)scale);
+    } else if (dh <= 128) {
+        hipLaunchKernelGGL((qkv_attention_kernel<128>), grid, dim3(256), 0,
+                           pe_current_stream(), in, out, B, S, H, dh,
+                           (float)scale);
+    } else {
+        TORCH_CHECK(false, "attention: head dim > 128 unsupported");
+    }
+    HIP_CHECK_LAST();
+    return ctx;
+}

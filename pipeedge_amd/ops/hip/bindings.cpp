@@ -1,0 +1,62 @@
+// Python bindings for the pipeedge-amd gfx950 kernel extension.
+//
+// The Python-side dispatch (pipeedge_amd/ops/__init__.py) calls these for
+// every GPU tensor; there is no eager fallback on GPU (fail-loud contract).
+#include <torch/extension.h>
+
+torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
+                        c10::optional<torch::Tensor> bias, int64_t act,
+                        c10::optional<torch::Tensor> residual2d);
+torch::Tensor debug_mfma_tile(torch::Tensor a, torch::Tensor b);
+torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor weight,
+                             torch::Tensor bias, double eps);
+torch::Tensor add_layernorm_bf16(torch::Tensor x, torch::Tensor residual,
+                                 torch::Tensor weight, torch::Tensor bias,
+                                 double eps);
+torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
+                                 double scale);
+torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
+
+namespace {
+
+// reshape [.., K] -> [M, K] for the GEMM, restoring leading dims after
+torch::Tensor linear_op(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias, int64_t act,
+                        c10::optional<torch::Tensor> residual) {
+    auto sizes = x.sizes().vec();
+    const long K = sizes.back();
+    auto x2d = x.reshape({-1, K});
+    c10::optional<torch::Tensor> res2d;
+    if (residual.has_value())
+        res2d = residual->reshape({-1, w.size(1)});
+    auto out = gemm_bf16(x2d, w, bias, act, res2d);
+    sizes.back() = w.size(1);
+    return out.reshape(sizes);
+}
+
+torch::Tensor qkv_attention_op(torch::Tensor qkv, int64_t num_heads,
+                               double scale,
+                               c10::optional<torch::Tensor> attn_mask) {
+    TORCH_CHECK(!attn_mask.has_value(),
+                "HIP attention: attn_mask not supported yet");
+    return qkv_attention_bf16(qkv, num_heads, scale);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("linear", &linear_op, "fused GEMM+bias+act+residual (bf16 MFMA)",
+          py::arg("x"), py::arg("weight"), py::arg("bias") = py::none(),
+          py::arg("act") = 0, py::arg("residual") = py::none());
+    m.def("layernorm", &layernorm_bf16, py::arg("x"), py::arg("weight"),
+          py::arg("bias"), py::arg("eps") = 1e-6);
+    m.def("add_layernorm", &add_layernorm_bf16, py::arg("x"),
+          py::arg("residual"), py::arg("weight"), py::arg("bias"),
+          py::arg("eps") = 1e-6);
+    m.def("qkv_attention", &qkv_attention_op, py::arg("qkv"),
+          py::arg("num_heads"), py::arg("scale"),
+          py::arg("attn_mask") = py::none());
+    m.def("patchify", &patchify_bf16, py::arg("pixels"),
+          py::arg("patch_size"));
+    m.def("_debug_mfma_tile", &debug_mfma_tile);
+}

@@ -1,0 +1,74 @@
+// Shared helpers for the pipeedge-amd CDNA4 (gfx950) kernels.
+//
+// Conventions:
+// - compute dtype bf16, accumulate f32 via MFMA (v_mfma_f32_16x16x32_bf16);
+// - wave = 64 lanes; blocks are multiples of 64 threads;
+// - fragment layout for mfma_f32_16x16x32_bf16 (verified on HW by
+//   tests/test_gpu_ops.py::test_mfma_fragment_layout):
+//     A[16r][32k]: lane l holds 8 contiguous k: row=l&15, k=(l>>4)*8+j
+//     B[32k][16c]: lane l holds 8 contiguous k: col=l&15, k=(l>>4)*8+j
+//     C/D[16r][16c]: lane l, reg r: col=l&15, row=(l>>4)*4+r
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#ifdef __HIPCC__
+#include <c10/hip/HIPStream.h>
+inline hipStream_t pe_current_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+#endif
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+
+__device__ __forceinline__ f32x4 mfma_16x16x32_bf16(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+__device__ __forceinline__ float bf2f(__bf16 v) {
+    return static_cast<float>(v);
+}
+
+__device__ __forceinline__ __bf16 f2bf(float v) {
+    return static_cast<__bf16>(v);
+}
+
+// exact GeLU (erf form) matching torch.nn.functional.gelu default
+__device__ __forceinline__ float gelu_f(float x) {
+    return 0.5f * x * (1.0f + erff(x * 0.70710678118654752f));
+}
+
+// activation codes shared with pipeedge_amd/ops/__init__.py (_ACT_CODE)
+#define ACT_NONE 0
+#define ACT_GELU 1
+#define ACT_TANH 2
+
+// reduce over the 16-lane groups that share a C-fragment row set
+// (lanes l and l^m for m in {1,2,4,8} have the same l>>4)
+template <typename Op>
+__device__ __forceinline__ float group16_reduce(float v, Op op) {
+    v = op(v, __shfl_xor(v, 1));
+    v = op(v, __shfl_xor(v, 2));
+    v = op(v, __shfl_xor(v, 4));
+    v = op(v, __shfl_xor(v, 8));
+    return v;
+}
+
+struct MaxOp {
+    __device__ float operator()(float a, float b) const { return fmaxf(a, b); }
+};
+struct SumOp {
+    __device__ float operator()(float a, float b) const { return a + b; }
+};
+
+#define HIP_CHECK_LAST()                                                     \
+    do {                                                                     \
+        hipError_t e_ = hipGetLastError();                                   \
+        if (e_ != hipSuccess) {                                              \
+            TORCH_CHECK(false, "HIP kernel launch failed: ",                 \
+                        hipGetErrorString(e_));                              \
+        }                                                                    \
+    } while (0)
