@@ -5,7 +5,7 @@ Capability parity with reference ``src/pipeedge/models/transformers/bert.py``:
 (reference bert.py:41-52), first shard owns embeddings (:76-80), last owns the
 pooler (:98-102); ``BertShardForSequenceClassification`` strips the ``bert.``
 key prefix of fine-tuned checkpoints (:191-196). HF state_dict npz key layout
-(``[N, K]`` weights, transposed into our GEMM-ready ``[K, N]`` at load).
+(``[N, K]`` weights == our native MFMA GEMM layout: loaded directly).
 
 Sublayer semantics (post-LN, reference bert.py:41-52):
   0: self-attention                      -> (context, hidden_in)
@@ -71,7 +71,7 @@ class BertLayerShard(ModuleShard):
         i = config.intermediate_size
         self.scale = 1.0 / math.sqrt(config.head_dim)
         if self.has_layer(0):
-            self.qkv_weight = _param((d, 3 * d))
+            self.qkv_weight = _param((3 * d, d))
             self.qkv_bias = _param((3 * d,))
         if self.has_layer(1):
             self.out_weight = _param((d, d))
@@ -79,10 +79,10 @@ class BertLayerShard(ModuleShard):
             self.out_ln_weight = _param((d,))
             self.out_ln_bias = _param((d,))
         if self.has_layer(2):
-            self.fc1_weight = _param((d, i))
+            self.fc1_weight = _param((i, d))
             self.fc1_bias = _param((i,))
         if self.has_layer(3):
-            self.fc2_weight = _param((i, d))
+            self.fc2_weight = _param((d, i))
             self.fc2_bias = _param((d,))
             self.fc2_ln_weight = _param((d,))
             self.fc2_ln_bias = _param((d,))
@@ -165,28 +165,28 @@ class BertModelShard(ModuleShard):
         for layer in self.layers:
             root = f"encoder.layer.{layer.block_id}."
             if layer.has_layer(0):
-                # HF [N, K] weights -> [K, N]
-                qkv_w = [_t(weights[root + f"attention.self.{n}.weight"]).t()
+                # HF [N, K] weights: direct
+                qkv_w = [_t(weights[root + f"attention.self.{n}.weight"])
                          for n in ("query", "key", "value")]
-                layer.qkv_weight.copy_(torch.cat(qkv_w, dim=1))
+                layer.qkv_weight.copy_(torch.cat(qkv_w, dim=0))
                 qkv_b = [_t(weights[root + f"attention.self.{n}.bias"])
                          for n in ("query", "key", "value")]
                 layer.qkv_bias.copy_(torch.cat(qkv_b))
             if layer.has_layer(1):
-                layer.out_weight.copy_(_t(weights[root + "attention.output.dense.weight"]).t())
+                layer.out_weight.copy_(_t(weights[root + "attention.output.dense.weight"]))
                 layer.out_bias.copy_(_t(weights[root + "attention.output.dense.bias"]))
                 layer.out_ln_weight.copy_(_t(weights[root + "attention.output.LayerNorm.weight"]))
                 layer.out_ln_bias.copy_(_t(weights[root + "attention.output.LayerNorm.bias"]))
             if layer.has_layer(2):
-                layer.fc1_weight.copy_(_t(weights[root + "intermediate.dense.weight"]).t())
+                layer.fc1_weight.copy_(_t(weights[root + "intermediate.dense.weight"]))
                 layer.fc1_bias.copy_(_t(weights[root + "intermediate.dense.bias"]))
             if layer.has_layer(3):
-                layer.fc2_weight.copy_(_t(weights[root + "output.dense.weight"]).t())
+                layer.fc2_weight.copy_(_t(weights[root + "output.dense.weight"]))
                 layer.fc2_bias.copy_(_t(weights[root + "output.dense.bias"]))
                 layer.fc2_ln_weight.copy_(_t(weights[root + "output.LayerNorm.weight"]))
                 layer.fc2_ln_bias.copy_(_t(weights[root + "output.LayerNorm.bias"]))
         if self.shard_config.is_last:
-            self.pooler_weight.copy_(_t(weights["pooler.dense.weight"]).t())
+            self.pooler_weight.copy_(_t(weights["pooler.dense.weight"]))
             self.pooler_bias.copy_(_t(weights["pooler.dense.bias"]))
 
     @torch.no_grad()
@@ -222,7 +222,7 @@ class BertShardForSequenceClassification(ModuleShard):
                             if k.startswith("bert.")}
         self.bert = BertModelShard(config, shard_config, bert_weights)
         if shard_config.is_last:
-            self.classifier_weight = _param((config.hidden_size, config.num_labels))
+            self.classifier_weight = _param((config.num_labels, config.hidden_size))
             self.classifier_bias = _param((config.num_labels,))
             if model_weights is None:
                 init_param(self.classifier_weight, f"{config.name}/classifier/weight")
@@ -230,7 +230,7 @@ class BertShardForSequenceClassification(ModuleShard):
                            "zeros")
             else:
                 with torch.no_grad():
-                    self.classifier_weight.copy_(_t(model_weights["classifier.weight"]).t())
+                    self.classifier_weight.copy_(_t(model_weights["classifier.weight"]))
                     self.classifier_bias.copy_(_t(model_weights["classifier.bias"]))
 
     @torch.no_grad()

@@ -34,7 +34,7 @@ class DeiTEmbeddings(nn.Module):
         self.config = config
         d = config.hidden_size
         pdim = config.num_channels * config.patch_size ** 2
-        self.patch_weight = _param((pdim, d))
+        self.patch_weight = _param((d, pdim))
         self.patch_bias = _param((d,))
         self.cls_token = _param((1, 1, d))
         self.dist_token = _param((1, 1, d))
@@ -74,8 +74,8 @@ class DeiTModelShard(ViTModelShard):
             else:
                 emb.dist_token.zero_()
             emb.pos_embed.copy_(_t(weights["pos_embed"]))
-            # timm conv weight is [D, C, P, P]; our layout is [(c,ph,pw), D]
-            kern = weights["patch_embed.proj.weight"].transpose(1, 2, 3, 0)
+            # timm conv weight is [D, C, P, P] == [N, (c,ph,pw)]: direct
+            kern = weights["patch_embed.proj.weight"]
             emb.patch_weight.copy_(_t(kern).reshape(emb.patch_weight.shape))
             emb.patch_bias.copy_(_t(weights["patch_embed.proj.bias"]))
         for layer in self.layers:
@@ -83,19 +83,19 @@ class DeiTModelShard(ViTModelShard):
             if layer.has_layer(0):
                 layer.ln1_weight.copy_(_t(weights[root + "norm1.weight"]))
                 layer.ln1_bias.copy_(_t(weights[root + "norm1.bias"]))
-                # timm fused qkv weight is [3D, D] ([N, K]); ours is [D, 3D]
-                layer.qkv_weight.copy_(_t(weights[root + "attn.qkv.weight"]).t())
+                # timm fused qkv weight is [3D, D] == our [N, K]: direct
+                layer.qkv_weight.copy_(_t(weights[root + "attn.qkv.weight"]))
                 layer.qkv_bias.copy_(_t(weights[root + "attn.qkv.bias"]))
             if layer.has_layer(1):
-                layer.out_weight.copy_(_t(weights[root + "attn.proj.weight"]).t())
+                layer.out_weight.copy_(_t(weights[root + "attn.proj.weight"]))
                 layer.out_bias.copy_(_t(weights[root + "attn.proj.bias"]))
             if layer.has_layer(2):
                 layer.ln2_weight.copy_(_t(weights[root + "norm2.weight"]))
                 layer.ln2_bias.copy_(_t(weights[root + "norm2.bias"]))
-                layer.fc1_weight.copy_(_t(weights[root + "mlp.fc1.weight"]).t())
+                layer.fc1_weight.copy_(_t(weights[root + "mlp.fc1.weight"]))
                 layer.fc1_bias.copy_(_t(weights[root + "mlp.fc1.bias"]))
             if layer.has_layer(3):
-                layer.fc2_weight.copy_(_t(weights[root + "mlp.fc2.weight"]).t())
+                layer.fc2_weight.copy_(_t(weights[root + "mlp.fc2.weight"]))
                 layer.fc2_bias.copy_(_t(weights[root + "mlp.fc2.bias"]))
         if self.shard_config.is_last:
             self.layernorm_weight.copy_(_t(weights["norm.weight"]))
@@ -116,14 +116,14 @@ class DeiTShardForImageClassification(ViTShardForImageClassification):
         self.classifier_weight = None
         self.vit = DeiTModelShard(config, shard_config, model_weights)
         if shard_config.is_last and config.num_labels > 0:
-            self.classifier_weight = _param((config.hidden_size, config.num_labels))
+            self.classifier_weight = _param((config.num_labels, config.hidden_size))
             self.classifier_bias = _param((config.num_labels,))
             if model_weights is None:
                 init_param(self.classifier_weight, f"{config.name}/head/weight")
                 init_param(self.classifier_bias, f"{config.name}/head/bias", "zeros")
             else:
                 with torch.no_grad():
-                    self.classifier_weight.copy_(_t(model_weights["head.weight"]).t())
+                    self.classifier_weight.copy_(_t(model_weights["head.weight"]))
                     self.classifier_bias.copy_(_t(model_weights["head.bias"]))
 
     @staticmethod

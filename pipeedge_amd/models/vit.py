@@ -6,9 +6,10 @@ the same Google ``.npz`` checkpoint key layout (reference vit.py:120-159), and
 the same shard classes (`ViTModelShard`, `ViTShardForImageClassification`).
 
 Design differences (deliberate, MI355X-first):
-- no HF modules: parameters are held directly in GEMM-ready ``[K, N]`` layout
-  (the Google npz kernel layout - zero transposes at load) and all compute is
-  routed through ``pipeedge_amd.ops`` (HIP kernels on gfx950, torch on CPU);
+- no HF modules: parameters are held directly in the MFMA-native ``[N, K]``
+  GEMM layout (k-contiguous rows for both operands - see
+  pipeedge_amd/ops/torch_ref.py) and all compute is routed through
+  ``pipeedge_amd.ops`` (HIP kernels on gfx950, torch on CPU);
 - QKV is packed into one ``[D, 3D]`` weight so self-attention is a single
   GEMM + one fused attention kernel;
 - missing weights file => deterministic random init (synthetic benchmarking -
@@ -69,7 +70,7 @@ class ViTEmbeddings(nn.Module):
     """Patch projection + cls token + position embeddings.
 
     The 16x16/s16 conv is expressed as patchify (pure data movement) + GEMM,
-    which is the shape MFMA wants; weight layout ``[C*P*P, D]`` with patch
+    which is the shape MFMA wants; weight layout ``[D, C*P*P]`` with patch
     pixels ordered (c, ph, pw).
     """
 
@@ -78,7 +79,7 @@ class ViTEmbeddings(nn.Module):
         self.config = config
         d = config.hidden_size
         pdim = config.num_channels * config.patch_size ** 2
-        self.patch_weight = _param((pdim, d))
+        self.patch_weight = _param((d, pdim))
         self.patch_bias = _param((d,))
         self.cls_token = _param((1, 1, d))
         self.pos_embed = _param((1, config.seq_len, d))
@@ -107,7 +108,7 @@ class ViTLayerShard(ModuleShard):
         if self.has_layer(0):
             self.ln1_weight = _param((d,))
             self.ln1_bias = _param((d,))
-            self.qkv_weight = _param((d, 3 * d))
+            self.qkv_weight = _param((3 * d, d))
             self.qkv_bias = _param((3 * d,))
         if self.has_layer(1):
             self.out_weight = _param((d, d))
@@ -115,10 +116,10 @@ class ViTLayerShard(ModuleShard):
         if self.has_layer(2):
             self.ln2_weight = _param((d,))
             self.ln2_bias = _param((d,))
-            self.fc1_weight = _param((d, i))
+            self.fc1_weight = _param((i, d))
             self.fc1_bias = _param((i,))
         if self.has_layer(3):
-            self.fc2_weight = _param((i, d))
+            self.fc2_weight = _param((d, i))
             self.fc2_bias = _param((d,))
 
     def init_random(self, prefix: str):
@@ -202,8 +203,8 @@ class ViTModelShard(ModuleShard):
             emb = self.embeddings
             emb.cls_token.copy_(_t(weights["cls"]))
             emb.pos_embed.copy_(_t(weights["Transformer/posembed_input/pos_embedding"]))
-            # npz conv kernel is [P, P, C, D]; our patch vectors are (c, ph, pw)
-            kern = weights["embedding/kernel"].transpose(2, 0, 1, 3)
+            # npz conv kernel is [P, P, C, D]; ours is [D, (c, ph, pw)]
+            kern = weights["embedding/kernel"].transpose(3, 2, 0, 1)
             emb.patch_weight.copy_(_t(kern).reshape(emb.patch_weight.shape))
             emb.patch_bias.copy_(_t(weights["embedding/bias"]))
         d = self.config.hidden_size
@@ -213,23 +214,23 @@ class ViTModelShard(ModuleShard):
             if layer.has_layer(0):
                 layer.ln1_weight.copy_(_t(weights[root + "LayerNorm_0/scale"]))
                 layer.ln1_bias.copy_(_t(weights[root + "LayerNorm_0/bias"]))
-                # npz kernels are [D, H, dh] == [K, N] once flattened: no transpose
-                qkv_w = [_t(weights[attn + f"{n}/kernel"]).reshape(d, d)
+                # npz kernels are [D(in), H, dh] -> transpose to [N, K]
+                qkv_w = [_t(weights[attn + f"{n}/kernel"]).reshape(d, d).t()
                          for n in ("query", "key", "value")]
-                layer.qkv_weight.copy_(torch.cat(qkv_w, dim=1))
+                layer.qkv_weight.copy_(torch.cat(qkv_w, dim=0))
                 qkv_b = [_t(weights[attn + f"{n}/bias"]).reshape(-1)
                          for n in ("query", "key", "value")]
                 layer.qkv_bias.copy_(torch.cat(qkv_b))
             if layer.has_layer(1):
-                layer.out_weight.copy_(_t(weights[attn + "out/kernel"]).reshape(d, d))
+                layer.out_weight.copy_(_t(weights[attn + "out/kernel"]).reshape(d, d).t())
                 layer.out_bias.copy_(_t(weights[attn + "out/bias"]).reshape(-1))
             if layer.has_layer(2):
                 layer.ln2_weight.copy_(_t(weights[root + "LayerNorm_2/scale"]))
                 layer.ln2_bias.copy_(_t(weights[root + "LayerNorm_2/bias"]))
-                layer.fc1_weight.copy_(_t(weights[root + "MlpBlock_3/Dense_0/kernel"]))
+                layer.fc1_weight.copy_(_t(weights[root + "MlpBlock_3/Dense_0/kernel"]).t())
                 layer.fc1_bias.copy_(_t(weights[root + "MlpBlock_3/Dense_0/bias"]))
             if layer.has_layer(3):
-                layer.fc2_weight.copy_(_t(weights[root + "MlpBlock_3/Dense_1/kernel"]))
+                layer.fc2_weight.copy_(_t(weights[root + "MlpBlock_3/Dense_1/kernel"]).t())
                 layer.fc2_bias.copy_(_t(weights[root + "MlpBlock_3/Dense_1/bias"]))
         if self.shard_config.is_last:
             self.layernorm_weight.copy_(_t(weights["Transformer/encoder_norm/scale"]))
@@ -260,15 +261,15 @@ class ViTShardForImageClassification(ModuleShard):
             model_weights = open_npz(model_weights)
         self.vit = self.model_shard_cls(config, shard_config, model_weights)
         if shard_config.is_last and config.num_labels > 0:
-            self.classifier_weight = _param((config.hidden_size, config.num_labels))
+            self.classifier_weight = _param((config.num_labels, config.hidden_size))
             self.classifier_bias = _param((config.num_labels,))
             if model_weights is None:
                 init_param(self.classifier_weight, f"{config.name}/head/weight")
                 init_param(self.classifier_bias, f"{config.name}/head/bias", "zeros")
             else:
                 with torch.no_grad():
-                    # npz head/kernel is [D, num_labels] == [K, N]
-                    self.classifier_weight.copy_(_t(model_weights["head/kernel"]))
+                    # npz head/kernel is [D, num_labels] -> [N, K]
+                    self.classifier_weight.copy_(_t(model_weights["head/kernel"]).t())
                     self.classifier_bias.copy_(_t(model_weights["head/bias"]))
 
     @torch.no_grad()

@@ -76,11 +76,11 @@ def add_layernorm(x: Tensor, residual: Tensor, weight: Tensor, bias: Tensor,
 _ACT_CODE = {None: 0, "gelu": 1, "tanh": 2}
 
 
-def linear(x: Tensor, weight_kn: Tensor, bias: Optional[Tensor] = None,
+def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
            act: Optional[str] = None, residual: Optional[Tensor] = None) -> Tensor:
     if _use_hip(x):
-        return _hip.linear(x, weight_kn, bias, _ACT_CODE[act], residual)
-    return torch_ref.linear(x, weight_kn, bias, act, residual)
+        return _hip.linear(x, weight_nk, bias, _ACT_CODE[act], residual)
+    return torch_ref.linear(x, weight_nk, bias, act, residual)
 
 
 def attention_core(q: Tensor, k: Tensor, v: Tensor, scale: Optional[float] = None,

@@ -1,14 +1,20 @@
 // bf16 MFMA GEMM with fused epilogue (bias + activation + residual add).
 //
-// Computes C[M,N] = act(A[M,K] @ B[K,N] + bias) [+ residual], all bf16
-// row-major with f32 accumulation. B is stored [K,N] (the framework's
-// GEMM-ready weight layout; see pipeedge_amd/ops/torch_ref.py docstring).
+// Computes C[M,N] = act(A[M,K] @ B^T + bias) [+ residual] with B stored
+// [N,K] - the framework's native weight layout. [N,K] makes BOTH operands
+// k-contiguous per row, so A- and B-tiles stage identically.
 //
-// Structure (cdna_hip_programming.md section 5 canonical anatomy): 64x64x64
-// block tile, 4 waves in 2x2, each wave one 32x32 sub-tile of 2x2 MFMA
-// fragments; A and B^T tiles staged in padded LDS; fragment reads are
-// 16-B ds_read_b128. Replaces what the reference ran through HF Linear
-// modules on CPU (reference vit.py:45-53).
+// Two paths (cdna_hip_programming.md section 5):
+// - glds path (K % 64 == 0, the transformer shapes): 64x64x64 tiles staged
+//   with 16-B `global_load_lds` directly into double-buffered LDS
+//   (no VGPR round trip - Common-mistake #1), XOR-swizzled via the SOURCE
+//   address (rule 21) so ds_read_b128 fragment reads are low-conflict (T2),
+//   2-phase pipeline: next tile's glds issued before the current tile's
+//   MFMAs (T3 minimum recipe);
+// - generic path (any K): plain vectorized staging into padded LDS.
+//
+// Replaces what the reference ran through HF Linear modules on CPU
+// (reference vit.py:45-53).
 #include <torch/extension.h>
 
 #include "common.h"
@@ -18,107 +24,192 @@ namespace {
 constexpr int BM = 64;
 constexpr int BN = 64;
 constexpr int BK = 64;
-constexpr int PAD = 8;  // LDS row pad: breaks the 128-B-row bank conflict
+constexpr int PAD = 8;
 
-__global__ __launch_bounds__(256) void gemm_bf16_kernel(
+__device__ __forceinline__ void glds_b128(const void* gsrc, void* lds_dst) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gsrc,
+        (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
+}
+
+// read a bf16x8 fragment from a [64][64] lane-linear tile with the
+// (row&7)<<4 byte XOR swizzle (matches the staging source permutation)
+__device__ __forceinline__ bf16x8 lds_read_swz(const __bf16* tile, int row,
+                                               int kelem) {
+    const int byte = (row * 128 + kelem * 2) ^ ((row & 7) << 4);
+    return *reinterpret_cast<const bf16x8*>(
+        reinterpret_cast<const char*>(tile) + byte);
+}
+
+struct EpilogueArgs {
+    const __bf16* bias;
+    const __bf16* residual;
+    __bf16* C;
+    int M, N, act;
+};
+
+__device__ __forceinline__ void gemm_epilogue(
+        const f32x4 (&acc)[2][2], const EpilogueArgs& ep, int bm, int bn,
+        int wm, int wn, int lane) {
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+            const int col = bn + wn + ni * 16 + (lane & 15);
+            if (col >= ep.N) continue;
+            const float bv = ep.bias != nullptr ? bf2f(ep.bias[col]) : 0.0f;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
+                if (row >= ep.M) continue;
+                float v = acc[mi][ni][r] + bv;
+                if (ep.act == ACT_GELU) v = gelu_f(v);
+                else if (ep.act == ACT_TANH) v = tanhf(v);
+                if (ep.residual != nullptr)
+                    v += bf2f(ep.residual[(long)row * ep.N + col]);
+                ep.C[(long)row * ep.N + col] = f2bf(v);
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// glds path: K % 64 == 0
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
         __bf16* __restrict__ C, int M, int N, int K, int act) {
-    __shared__ __bf16 As[BM][BK + PAD];
-    __shared__ __bf16 Bs[BN][BK + PAD];  // B tile transposed: Bs[n][k]
+    // ONE shared object (guide section 5 trap 4a): [buf][A=0/B=1][64][64]
+    __shared__ __bf16 smem[2][2][BM][BK];
 
     const int bm = blockIdx.x * BM;
     const int bn = blockIdx.y * BN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wm = (wid >> 1) * 32;  // wave row offset in tile
-    const int wn = (wid & 1) * 32;   // wave col offset in tile
+    const int wm = (wid >> 1) * 32;
+    const int wn = (wid & 1) * 32;
+
+    // lane's role in staging: row-in-piece and swizzled source column
+    const int srow = lane >> 3;              // 0..7 within the 8-row piece
+    const int csrc = (lane & 7) ^ srow;      // XOR-swizzled 16-B column
+
+    auto stage = [&](int buf, int k0) {
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const int p = wid * 2 + i;       // piece 0..7 = rows p*8..p*8+7
+            const long arow = min(bm + p * 8 + srow, M - 1);
+            glds_b128(A + arow * K + k0 + csrc * 8,
+                      &smem[buf][0][p * 8][0]);
+            const long brow = min(bn + p * 8 + srow, N - 1);
+            glds_b128(B + brow * K + k0 + csrc * 8,
+                      &smem[buf][1][p * 8][0]);
+        }
+    };
+
+    f32x4 acc[2][2] = {};
+    stage(0, 0);
+    __syncthreads();  // barrier waits vmcnt(0) while glds in flight
+
+    const int n_tiles = K / BK;
+    for (int t = 0; t < n_tiles; ++t) {
+        const int buf = t & 1;
+        if (t + 1 < n_tiles) stage(buf ^ 1, (t + 1) * BK);
+        #pragma unroll
+        for (int kk = 0; kk < BK; kk += 32) {
+            const int kgrp = kk + (lane >> 4) * 8;
+            bf16x8 af[2], bfr[2];
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                af[mi] = lds_read_swz(&smem[buf][0][0][0],
+                                      wm + mi * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+                bfr[ni] = lds_read_swz(&smem[buf][1][0][0],
+                                       wn + ni * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < 2; ++ni)
+                    acc[mi][ni] = mfma_16x16x32_bf16(af[mi], bfr[ni],
+                                                     acc[mi][ni]);
+        }
+        __syncthreads();
+    }
+
+    gemm_epilogue(acc, {bias, residual, C, M, N, act}, bm, bn, wm, wn, lane);
+}
+
+// ---------------------------------------------------------------------------
+// generic path: any K (padded LDS, plain vectorized staging)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
+        const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+        const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
+        __bf16* __restrict__ C, int M, int N, int K, int act) {
+    __shared__ __bf16 As[BM][BK + PAD];
+    __shared__ __bf16 Bs[BN][BK + PAD];
+
+    const int bm = blockIdx.x * BM;
+    const int bn = blockIdx.y * BN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wm = (wid >> 1) * 32;
+    const int wn = (wid & 1) * 32;
 
     f32x4 acc[2][2] = {};
 
     for (int k0 = 0; k0 < K; k0 += BK) {
-        // ---- stage A tile: each thread 2x 16-B pieces ----
         #pragma unroll
         for (int it = 0; it < 2; ++it) {
             const int r = (tid >> 3) + it * 32;
             const int c = (tid & 7) * 8;
-            const int gr = bm + r;
             const int gc = k0 + c;
-            bf16x8 v = {};
-            if (gr < M) {
-                if (gc + 7 < K) {
-                    v = *reinterpret_cast<const bf16x8*>(
-                        A + (long)gr * K + gc);
-                } else {
-                    for (int j = 0; j < 8; ++j)
-                        if (gc + j < K) v[j] = A[(long)gr * K + gc + j];
-                }
+            bf16x8 va = {}, vb = {};
+            if (bm + r < M && gc + 7 < K) {
+                va = *reinterpret_cast<const bf16x8*>(
+                    A + (long)(bm + r) * K + gc);
+            } else if (bm + r < M) {
+                for (int j = 0; j < 8; ++j)
+                    if (gc + j < K) va[j] = A[(long)(bm + r) * K + gc + j];
             }
-            *reinterpret_cast<bf16x8*>(&As[r][c]) = v;
-        }
-        // ---- stage B tile transposed ----
-        #pragma unroll
-        for (int it = 0; it < 2; ++it) {
-            const int kr = (tid >> 3) + it * 32;
-            const int nc = (tid & 7) * 8;
-            const int gk = k0 + kr;
-            const int gn = bn + nc;
-            bf16x8 v = {};
-            if (gk < K) {
-                if (gn + 7 < N) {
-                    v = *reinterpret_cast<const bf16x8*>(
-                        B + (long)gk * N + gn);
-                } else {
-                    for (int j = 0; j < 8; ++j)
-                        if (gn + j < N) v[j] = B[(long)gk * N + gn + j];
-                }
+            if (bn + r < N && gc + 7 < K) {
+                vb = *reinterpret_cast<const bf16x8*>(
+                    B + (long)(bn + r) * K + gc);
+            } else if (bn + r < N) {
+                for (int j = 0; j < 8; ++j)
+                    if (gc + j < K) vb[j] = B[(long)(bn + r) * K + gc + j];
             }
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) Bs[nc + j][kr] = v[j];
+            *reinterpret_cast<bf16x8*>(&As[r][c]) = va;
+            *reinterpret_cast<bf16x8*>(&Bs[r][c]) = vb;
         }
         __syncthreads();
 
-        // ---- MFMA inner loop over the K-tile ----
         #pragma unroll
         for (int kk = 0; kk < BK; kk += 32) {
-            const int kbase = kk + (lane >> 4) * 8;
+            const int kgrp = kk + (lane >> 4) * 8;
+            bf16x8 af[2], bfr[2];
             #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
-                const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                    &As[wm + mi * 16 + (lane & 15)][kbase]);
+            for (int mi = 0; mi < 2; ++mi)
+                af[mi] = *reinterpret_cast<const bf16x8*>(
+                    &As[wm + mi * 16 + (lane & 15)][kgrp]);
+            #pragma unroll
+            for (int ni = 0; ni < 2; ++ni)
+                bfr[ni] = *reinterpret_cast<const bf16x8*>(
+                    &Bs[wn + ni * 16 + (lane & 15)][kgrp]);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
                 #pragma unroll
-                for (int ni = 0; ni < 2; ++ni) {
-                    const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-                        &Bs[wn + ni * 16 + (lane & 15)][kbase]);
-                    acc[mi][ni] = mfma_16x16x32_bf16(a, b, acc[mi][ni]);
-                }
-            }
+                for (int ni = 0; ni < 2; ++ni)
+                    acc[mi][ni] = mfma_16x16x32_bf16(af[mi], bfr[ni],
+                                                     acc[mi][ni]);
         }
         __syncthreads();
     }
 
-    // ---- epilogue: bias + activation + residual, bf16 store ----
-    #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
-        #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-            const int col = bn + wn + ni * 16 + (lane & 15);
-            if (col >= N) continue;
-            const float bv = bias != nullptr ? bf2f(bias[col]) : 0.0f;
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
-                if (row >= M) continue;
-                float v = acc[mi][ni][r] + bv;
-                if (act == ACT_GELU) v = gelu_f(v);
-                else if (act == ACT_TANH) v = tanhf(v);
-                if (residual != nullptr)
-                    v += bf2f(residual[(long)row * N + col]);
-                C[(long)row * N + col] = f2bf(v);
-            }
-        }
-    }
+    gemm_epilogue(acc, {bias, residual, C, M, N, act}, bm, bn, wm, wn, lane);
 }
 
 }  // namespace
@@ -130,8 +221,8 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                 "gemm: A must be bf16 on GPU");
     TORCH_CHECK(b.dtype() == torch::kBFloat16, "gemm: B must be bf16");
     TORCH_CHECK(a2d.dim() == 2 && b.dim() == 2);
-    const int M = a2d.size(0), K = a2d.size(1), N = b.size(1);
-    TORCH_CHECK(b.size(0) == K, "gemm: inner dims mismatch");
+    const int M = a2d.size(0), K = a2d.size(1), N = b.size(0);
+    TORCH_CHECK(b.size(1) == K, "gemm: inner dims mismatch (B is [N,K])");
     auto a_c = a2d.contiguous();
     auto b_c = b.contiguous();
     auto out = torch::empty({M, N}, a2d.options());
@@ -151,13 +242,18 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
         res_p = reinterpret_cast<const __bf16*>(res_c.data_ptr());
     }
     dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-    hipLaunchKernelGGL(gemm_bf16_kernel, grid, dim3(256), 0,
-                       pe_current_stream(),
-                       reinterpret_cast<const __bf16*>(a_c.data_ptr()),
-                       reinterpret_cast<const __bf16*>(b_c.data_ptr()),
-                       bias_p, res_p,
-                       reinterpret_cast<__bf16*>(out.data_ptr()),
-                       M, N, K, (int)act);
+    const auto* ap = reinterpret_cast<const __bf16*>(a_c.data_ptr());
+    const auto* bp = reinterpret_cast<const __bf16*>(b_c.data_ptr());
+    auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
+    if (K % BK == 0) {
+        hipLaunchKernelGGL(gemm_bf16_glds_kernel, grid, dim3(256), 0,
+                           pe_current_stream(), ap, bp, bias_p, res_p, cp,
+                           M, N, K, (int)act);
+    } else {
+        hipLaunchKernelGGL(gemm_bf16_generic_kernel, grid, dim3(256), 0,
+                           pe_current_stream(), ap, bp, bias_p, res_p, cp,
+                           M, N, K, (int)act);
+    }
     HIP_CHECK_LAST();
     return out;
 }
@@ -193,8 +289,7 @@ torch::Tensor debug_mfma_tile(torch::Tensor a, torch::Tensor b) {
     TORCH_CHECK(b.sizes() == torch::IntArrayRef({32, 16}));
     auto a_c = a.to(torch::kBFloat16).contiguous();
     auto b_c = b.to(torch::kBFloat16).contiguous();
-    auto out = torch::empty({16, 16},
-                            a.options().dtype(torch::kFloat32));
+    auto out = torch::empty({16, 16}, a.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(debug_mfma_kernel, dim3(1), dim3(64), 0,
                        pe_current_stream(),
                        reinterpret_cast<const __bf16*>(a_c.data_ptr()),

@@ -6,10 +6,11 @@ run in fp32) and the execution path on CPU, where the reference framework ran
 its shards eagerly through HF modules (reference vit.py:55-70, bert.py:41-52).
 
 Weight layout convention (MI355X-native): all linear weights are stored
-``[K, N]`` (in-features x out-features), i.e. GEMM-ready ``y = x @ w + b``.
-This is also the layout of the Google ViT ``.npz`` checkpoints, so the
-checkpoint contract (reference vit.py:120-159) is a direct copy for ViT.
-(torch.nn.Linear stores ``[N, K]`` instead and transposes on every call.)
+``[N, K]`` (out-features x in-features, the torch.nn.Linear layout). This is
+the MFMA-optimal operand layout on gfx950: both GEMM operands are then
+k-contiguous per row, so A- and B-fragments alike stage through LDS with
+16-B `global_load_lds` pieces and read back as single `ds_read_b128`s
+(cdna_hip_programming.md section 5).
 """
 import math
 from typing import Optional
@@ -29,13 +30,13 @@ def add_layernorm(x: Tensor, residual: Tensor, weight: Tensor, bias: Tensor,
     return layernorm(x + residual, weight, bias, eps)
 
 
-def linear(x: Tensor, weight_kn: Tensor, bias: Optional[Tensor] = None,
+def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
            act: Optional[str] = None, residual: Optional[Tensor] = None) -> Tensor:
     """GEMM with optionally fused bias + activation + residual add.
 
-    ``weight_kn`` is ``[K, N]``; ``y = act(x @ w + b) + residual``.
+    ``weight_nk`` is ``[N, K]``; ``y = act(x @ w.T + b) + residual``.
     """
-    y = x.matmul(weight_kn)
+    y = x.matmul(weight_nk.t())
     if bias is not None:
         y = y + bias
     if act == "gelu":

@@ -42,7 +42,7 @@ def test_mfma_fragment_layout():
 def test_gemm_matches_fp32(m, n, k):
     torch.manual_seed(m + n + k)
     a = torch.randn(m, k) * 0.5
-    b = torch.randn(k, n) * 0.1
+    b = torch.randn(n, k) * 0.1
     bias = torch.randn(n)
     ref = torch_ref.linear(a, b, bias)
     out = _hip().linear(a.bfloat16().cuda(), b.bfloat16().cuda(),
@@ -53,7 +53,7 @@ def test_gemm_matches_fp32(m, n, k):
 def test_gemm_gelu_residual():
     torch.manual_seed(0)
     a = torch.randn(197, 768) * 0.5
-    b = torch.randn(768, 768) * 0.05
+    b = torch.randn(768, 768) * 0.05  # square: [N,K]
     bias = torch.randn(768) * 0.1
     res = torch.randn(197, 768)
     ref = torch_ref.linear(a, b, bias, act="gelu", residual=res)
@@ -172,6 +172,6 @@ def test_ops_dispatch_uses_hip_on_gpu():
     eager fallback) - checked via the loaded module object."""
     assert ops.hip_available()
     x = torch.randn(4, 768).bfloat16().cuda()
-    w = torch.randn(768, 768).bfloat16().cuda() * 0.02
+    w = torch.randn(768, 768).bfloat16().cuda() * 0.02  # [N,K]
     y = ops.linear(x, w)
     assert y.dtype == torch.bfloat16 and y.is_cuda

@@ -62,7 +62,7 @@ def test_vit_npz_roundtrip(tmp_weights_dir):
     assert torch.equal(m1(x), m2(x))
     # weights actually loaded (not random-init): classifier comes from npz
     with np.load(path) as w:
-        assert np.allclose(m1.classifier_weight.numpy(), w["head/kernel"])
+        assert np.allclose(m1.classifier_weight.numpy(), w["head/kernel"].T)
 
 
 def test_deit_stitched_equals_full(tmp_weights_dir):
