@@ -28,9 +28,9 @@ torch::Tensor linear_op(torch::Tensor x, torch::Tensor w,
     auto x2d = x.reshape({-1, K});
     c10::optional<torch::Tensor> res2d;
     if (residual.has_value())
-        res2d = residual->reshape({-1, w.size(1)});
+        res2d = residual->reshape({-1, w.size(0)});  // w is [N, K]
     auto out = gemm_bf16(x2d, w, bias, act, res2d);
-    sizes.back() = w.size(1);
+    sizes.back() = w.size(0);
     return out.reshape(sizes);
 }
 
