@@ -6,7 +6,8 @@
 
 torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                         c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual2d);
+                        c10::optional<torch::Tensor> residual2d,
+                        int64_t tile = 0);
 torch::Tensor debug_mfma_tile(torch::Tensor a, torch::Tensor b);
 torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor weight,
                              torch::Tensor bias, double eps);
@@ -22,14 +23,15 @@ namespace {
 // reshape [.., K] -> [M, K] for the GEMM, restoring leading dims after
 torch::Tensor linear_op(torch::Tensor x, torch::Tensor w,
                         c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual) {
+                        c10::optional<torch::Tensor> residual,
+                        int64_t tile = 0) {
     auto sizes = x.sizes().vec();
     const long K = sizes.back();
     auto x2d = x.reshape({-1, K});
     c10::optional<torch::Tensor> res2d;
     if (residual.has_value())
         res2d = residual->reshape({-1, w.size(0)});  // w is [N, K]
-    auto out = gemm_bf16(x2d, w, bias, act, res2d);
+    auto out = gemm_bf16(x2d, w, bias, act, res2d, tile);
     sizes.back() = w.size(0);
     return out.reshape(sizes);
 }
@@ -47,7 +49,8 @@ torch::Tensor qkv_attention_op(torch::Tensor qkv, int64_t num_heads,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("linear", &linear_op, "fused GEMM+bias+act+residual (bf16 MFMA)",
           py::arg("x"), py::arg("weight"), py::arg("bias") = py::none(),
-          py::arg("act") = 0, py::arg("residual") = py::none());
+          py::arg("act") = 0, py::arg("residual") = py::none(),
+          py::arg("tile") = 0);
     m.def("layernorm", &layernorm_bf16, py::arg("x"), py::arg("weight"),
           py::arg("bias"), py::arg("eps") = 1e-6);
     m.def("add_layernorm", &add_layernorm_bf16, py::arg("x"),

@@ -48,13 +48,14 @@ struct EpilogueArgs {
     int M, N, act;
 };
 
+template <int MI, int NI>
 __device__ __forceinline__ void gemm_epilogue(
-        const f32x4 (&acc)[2][2], const EpilogueArgs& ep, int bm, int bn,
+        const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
         int wm, int wn, int lane) {
     #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+    for (int mi = 0; mi < MI; ++mi) {
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
+        for (int ni = 0; ni < NI; ++ni) {
             const int col = bn + wn + ni * 16 + (lane & 15);
             if (col >= ep.N) continue;
             const float bv = ep.bias != nullptr ? bf2f(ep.bias[col]) : 0.0f;
@@ -74,22 +75,27 @@ __device__ __forceinline__ void gemm_epilogue(
 }
 
 // ---------------------------------------------------------------------------
-// glds path: K % 64 == 0
+// glds path: K % 64 == 0. Tile = (32*MI) x (32*NI) with 4 waves in 2x2,
+// each wave an MIxNI grid of 16x16 fragments. <2,2> = 64x64 (small-N
+// shapes need the block count), <4,4> = 128x128 (fc1/qkv class).
 // ---------------------------------------------------------------------------
+template <int MI, int NI>
 __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
         __bf16* __restrict__ C, int M, int N, int K, int act) {
-    // ONE shared object (guide section 5 trap 4a): [buf][A=0/B=1][64][64]
-    __shared__ __bf16 smem[2][2][BM][BK];
+    constexpr int TBM = 32 * MI;
+    constexpr int TBN = 32 * NI;
+    // ONE shared object (guide section 5 trap 4a): [buf][TBM+TBN rows][64]
+    __shared__ __bf16 smem[2][TBM + TBN][BK];
 
-    const int bm = blockIdx.x * BM;
-    const int bn = blockIdx.y * BN;
+    const int bm = blockIdx.x * TBM;
+    const int bn = blockIdx.y * TBN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wm = (wid >> 1) * 32;
-    const int wn = (wid & 1) * 32;
+    const int wm = (wid >> 1) * (16 * MI);
+    const int wn = (wid & 1) * (16 * NI);
 
     // lane's role in staging: row-in-piece and swizzled source column
     const int srow = lane >> 3;              // 0..7 within the 8-row piece
@@ -97,18 +103,21 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
 
     auto stage = [&](int buf, int k0) {
         #pragma unroll
-        for (int i = 0; i < 2; ++i) {
-            const int p = wid * 2 + i;       // piece 0..7 = rows p*8..p*8+7
+        for (int i = 0; i < TBM / 32; ++i) {
+            const int p = wid * (TBM / 32) + i;  // rows p*8..p*8+7 of A
             const long arow = min(bm + p * 8 + srow, M - 1);
-            glds_b128(A + arow * K + k0 + csrc * 8,
-                      &smem[buf][0][p * 8][0]);
+            glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * 8][0]);
+        }
+        #pragma unroll
+        for (int i = 0; i < TBN / 32; ++i) {
+            const int p = wid * (TBN / 32) + i;  // rows p*8..p*8+7 of B
             const long brow = min(bn + p * 8 + srow, N - 1);
             glds_b128(B + brow * K + k0 + csrc * 8,
-                      &smem[buf][1][p * 8][0]);
+                      &smem[buf][TBM + p * 8][0]);
         }
     };
 
-    f32x4 acc[2][2] = {};
+    f32x4 acc[MI][NI] = {};
     stage(0, 0);
     __syncthreads();  // barrier waits vmcnt(0) while glds in flight
 
@@ -119,26 +128,27 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         #pragma unroll
         for (int kk = 0; kk < BK; kk += 32) {
             const int kgrp = kk + (lane >> 4) * 8;
-            bf16x8 af[2], bfr[2];
+            bf16x8 af[MI], bfr[NI];
             #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
-                af[mi] = lds_read_swz(&smem[buf][0][0][0],
+            for (int mi = 0; mi < MI; ++mi)
+                af[mi] = lds_read_swz(&smem[buf][0][0],
                                       wm + mi * 16 + (lane & 15), kgrp);
             #pragma unroll
-            for (int ni = 0; ni < 2; ++ni)
-                bfr[ni] = lds_read_swz(&smem[buf][1][0][0],
+            for (int ni = 0; ni < NI; ++ni)
+                bfr[ni] = lds_read_swz(&smem[buf][TBM][0],
                                        wn + ni * 16 + (lane & 15), kgrp);
             #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
                 #pragma unroll
-                for (int ni = 0; ni < 2; ++ni)
+                for (int ni = 0; ni < NI; ++ni)
                     acc[mi][ni] = mfma_16x16x32_bf16(af[mi], bfr[ni],
                                                      acc[mi][ni]);
         }
         __syncthreads();
     }
 
-    gemm_epilogue(acc, {bias, residual, C, M, N, act}, bm, bn, wm, wn, lane);
+    gemm_epilogue<MI, NI>(acc, {bias, residual, C, M, N, act}, bm, bn, wm,
+                          wn, lane);
 }
 
 // ---------------------------------------------------------------------------
@@ -209,14 +219,16 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         __syncthreads();
     }
 
-    gemm_epilogue(acc, {bias, residual, C, M, N, act}, bm, bn, wm, wn, lane);
+    gemm_epilogue<2, 2>(acc, {bias, residual, C, M, N, act}, bm, bn, wm,
+                        wn, lane);
 }
 
 }  // namespace
 
 torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                         c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual2d) {
+                        c10::optional<torch::Tensor> residual2d,
+                        int64_t tile) {
     TORCH_CHECK(a2d.is_cuda() && a2d.dtype() == torch::kBFloat16,
                 "gemm: A must be bf16 on GPU");
     TORCH_CHECK(b.dtype() == torch::kBFloat16, "gemm: B must be bf16");
@@ -241,15 +253,37 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                     "gemm: residual shape mismatch");
         res_p = reinterpret_cast<const __bf16*>(res_c.data_ptr());
     }
-    dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
     const auto* ap = reinterpret_cast<const __bf16*>(a_c.data_ptr());
     const auto* bp = reinterpret_cast<const __bf16*>(b_c.data_ptr());
     auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
     if (K % BK == 0) {
-        hipLaunchKernelGGL(gemm_bf16_glds_kernel, grid, dim3(256), 0,
-                           pe_current_stream(), ap, bp, bias_p, res_p, cp,
-                           M, N, K, (int)act);
+        // tile choice: empirically swept (tests/gpu sweep); override via
+        // the `tile` arg (0 = auto heuristic)
+        int t = (int)tile;
+        if (t == 0) {
+            // from the on-HW sweep (tools/gemm_sweep.py): 64x64 wins at
+            // transformer-inference shapes (grid occupancy bound); 128x64
+            // only for wide-N long-M (fc1 class)
+            t = (N >= 3072 && M >= 1024) ? 2 : 1;
+        }
+        if (t == 3) {
+            dim3 grid((M + 127) / 128, (N + 127) / 128);
+            hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 4>), grid,
+                               dim3(256), 0, pe_current_stream(), ap, bp,
+                               bias_p, res_p, cp, M, N, K, (int)act);
+        } else if (t == 2) {
+            dim3 grid((M + 127) / 128, (N + 63) / 64);
+            hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 2>), grid,
+                               dim3(256), 0, pe_current_stream(), ap, bp,
+                               bias_p, res_p, cp, M, N, K, (int)act);
+        } else {
+            dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+            hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2>), grid,
+                               dim3(256), 0, pe_current_stream(), ap, bp,
+                               bias_p, res_p, cp, M, N, K, (int)act);
+        }
     } else {
+        dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
         hipLaunchKernelGGL(gemm_bf16_generic_kernel, grid, dim3(256), 0,
                            pe_current_stream(), ap, bp, bias_p, res_p, cp,
                            M, N, K, (int)act);

@@ -11,12 +11,14 @@
 namespace {
 
 template <bool WITH_RESIDUAL>
-__global__ __launch_bounds__(64) void layernorm_kernel(
+__global__ __launch_bounds__(256) void layernorm_kernel(
         const __bf16* __restrict__ x, const __bf16* __restrict__ residual,
         const __bf16* __restrict__ weight, const __bf16* __restrict__ bias,
-        __bf16* __restrict__ out, int D, float eps) {
-    const long row = blockIdx.x;
-    const int lane = threadIdx.x;
+        __bf16* __restrict__ out, long rows, int D, float eps) {
+    // 4 waves per block, one row per wave (wave-local shfl reduction only)
+    const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= rows) return;
+    const int lane = threadIdx.x & 63;
     const __bf16* xr = x + row * D;
     const __bf16* rr = WITH_RESIDUAL ? residual + row * D : nullptr;
 
@@ -93,14 +95,14 @@ torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor weight,
     auto out = torch::empty_like(x_c);
     const int D = x.size(-1);
     const long rows = x.numel() / D;
-    hipLaunchKernelGGL((layernorm_kernel<false>), dim3(rows), dim3(64), 0,
-                       pe_current_stream(),
+    hipLaunchKernelGGL((layernorm_kernel<false>), dim3((rows + 3) / 4),
+                       dim3(256), 0, pe_current_stream(),
                        reinterpret_cast<const __bf16*>(x_c.data_ptr()),
                        nullptr,
                        reinterpret_cast<const __bf16*>(w_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(b_c.data_ptr()),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       D, (float)eps);
+                       rows, D, (float)eps);
     HIP_CHECK_LAST();
     return out;
 }
@@ -117,14 +119,14 @@ torch::Tensor add_layernorm_bf16(torch::Tensor x, torch::Tensor residual,
     auto out = torch::empty_like(x_c);
     const int D = x.size(-1);
     const long rows = x.numel() / D;
-    hipLaunchKernelGGL((layernorm_kernel<true>), dim3(rows), dim3(64), 0,
-                       pe_current_stream(),
+    hipLaunchKernelGGL((layernorm_kernel<true>), dim3((rows + 3) / 4),
+                       dim3(256), 0, pe_current_stream(),
                        reinterpret_cast<const __bf16*>(x_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(r_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(w_c.data_ptr()),
                        reinterpret_cast<const __bf16*>(b_c.data_ptr()),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
-                       D, (float)eps);
+                       rows, D, (float)eps);
     HIP_CHECK_LAST();
     return out;
 }
