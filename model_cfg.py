@@ -108,6 +108,35 @@ def module_shard_factory(model_name: str, model_file: Optional[str],
     return shard
 
 
+def _dist_rpc_pipeline_stage_factory(*args, **kwargs):
+    """Build a DistRpcPipelineStage on the globally-configured device
+    (reference model_cfg.py:98-103); runs on the stage's owner rank."""
+    from pipeedge_amd.comm import rpc as rpc_mod
+    stage = rpc_mod.DistRpcPipelineStage(*args, **kwargs)
+    return stage
+
+
+def dist_rpc_pipeline_factory(model_name: str, model_file: Optional[str],
+                              stage_ranks: List[int],
+                              stage_layers: List[Tuple[int, int]],
+                              results_to: int,
+                              results_cb: Callable[[Any], None]):
+    """Build an RPC pipeline with remotely-constructed stages
+    (reference model_cfg.py:103-126)."""
+    from torch.distributed import rpc as trpc
+    from pipeedge_amd.comm import rpc as rpc_mod
+    stage_rrefs = []
+    assert len(stage_ranks) == len(stage_layers) > 0
+    for i, (dst_rank, layers) in enumerate(zip(stage_ranks, stage_layers)):
+        rref = trpc.remote(
+            dst_rank, _dist_rpc_pipeline_stage_factory,
+            args=(module_shard_factory,
+                  (model_name, model_file, layers[0], layers[1], i)),
+            kwargs={"stage_id": i})
+        stage_rrefs.append(rref)
+    return rpc_mod.DistRpcPipeline(stage_rrefs, results_to, results_cb)
+
+
 def stage_wire_schemas(model_name: str, stage_layers: List[Tuple[int, int]],
                        ubatch_size: int, stage_quant: List[int],
                        dtype: torch.dtype):

@@ -29,3 +29,19 @@ def load_device_types_file(file_path: str) -> dict:
 def load_devices_file(file_path: str) -> dict:
     """devices.yml: map of device type name -> list of host names."""
     return yaml_load(file_path)
+
+
+# reference-compatible loader aliases (reference sched/yaml_files.py:15-49)
+def yaml_models_load(file_path: str) -> dict:
+    """Load a models.yml file."""
+    return yaml_load(file_path) or {}
+
+
+def yaml_device_types_load(file_path: str) -> dict:
+    """Load a device_types.yml file."""
+    return yaml_load(file_path) or {}
+
+
+def yaml_device_neighbors_world_load(file_path: str) -> dict:
+    """Load a device neighbors world file: {host: {neighbor: {bw_Mbps}}}."""
+    return yaml_load(file_path) or {}
