@@ -202,6 +202,16 @@ class MonitorContext:
     def get_count(self, key=None):
         return self._hbs[key or self._default_key].count()
 
+    def get_tag(self, key=None):
+        """Iteration counter for a key (next tag to be assigned)."""
+        return self._tags[key or self._default_key]
+
+    def get_window_size(self, key=None):
+        return self.window_size
+
+    def get_window_heartrate(self, key=None):
+        return self._hbs[key or self._default_key].get_heart_rate("window")
+
     def __getattr__(self, name):
         # get_{instant,window,global}_{metric} dynamic getters
         for scope in ("instant", "window", "global"):

@@ -158,13 +158,146 @@ def get_pipeline_sched(world_size: int, hosts: Optional[List[str]],
 # runtime.py:73-119; wire format unchanged)
 # ---------------------------------------------------------------------------
 
-def make_quant_send_transform(module, pad_words_per_tensor: List[int]):
-    """Encode stage outputs with the module's current (adaptive) quant_bit."""
+# ---------------------------------------------------------------------------
+# adaptive quantization policies (QuantPipe; parity: reference
+# runtime.py:121-216). Invoked per microbatch inside the quant send
+# transform - the reference registers them as module forward hooks, which is
+# the same call point (after the shard computes, before encoding).
+# ---------------------------------------------------------------------------
+
+from pipeedge_amd.quant import compression_factor  # noqa: E402
+from pipeedge_amd.utils import quant as quantutil  # noqa: E402
+
+# largest bitwidths in [2, 32] with unique discrete compression factors
+BITWIDTHS = [i for i in range(32, 1, -1)
+             if int(compression_factor(i)) > int(compression_factor(i + 1))]
+_MODULE_QUANT_CONTROLLERS = {}
+_MODULE_QUANT_CONTROLLERS_LOCK = threading.Lock()
+
+
+def set_quant_bandwidth_heuristic(module, outputs) -> None:
+    """Pick a bitwidth so window send work fits the rate constraint."""
+    with monitoring.get_locked_context(MONITORING_KEY_SEND) as mctx:
+        tag = mctx.get_tag(key=MONITORING_KEY_SEND)
+        window_size = mctx.get_window_size(key=MONITORING_KEY_SEND)
+        bandwidth = mctx.get_window_perf(key=MONITORING_KEY_SEND)
+        send_work = mctx.get_window_work(key=MONITORING_KEY_SEND)
+    if tag == 0 or tag % window_size != 0:
+        return
+    target_rate = module.rate_constraint.item()
+    if target_rate > 0:
+        ubatch_size = models.get_microbatch_size(outputs, verify=True)
+        target_time = ubatch_size * window_size / target_rate
+    else:
+        target_time = float("inf")
+    target_datasize = target_time * bandwidth
+    quant_bit = module.quant_bit.item()
+    if target_datasize <= 0:
+        return
+    if quant_bit > 0:
+        compress_ratio = int(send_work * (32 / quant_bit) / target_datasize) + 1
+    else:
+        compress_ratio = int(send_work / target_datasize) + 1
+    for limit, bit in ((1, 0), (2, 16), (4, 8), (5, 6), (8, 4)):
+        if compress_ratio <= limit:
+            module.quant_bit = torch.tensor(bit)
+            break
+    else:
+        module.quant_bit = torch.tensor(2)
+    logger.info("Adaptive quantization (heuristic): bitwidth=%d",
+                int(module.quant_bit))
+
+
+def set_quant_bandwidth_heuristic2(module, outputs) -> None:
+    """Solve the discrete-packing constraint directly."""
+    with monitoring.get_locked_context(MONITORING_KEY_SEND) as mctx:
+        tag = mctx.get_tag(key=MONITORING_KEY_SEND)
+        window_size = mctx.get_window_size(key=MONITORING_KEY_SEND)
+        bandwidth = mctx.get_window_perf(key=MONITORING_KEY_SEND)
+    if tag == 0 or tag % window_size != 0:
+        return
+    tensors = (outputs,) if isinstance(outputs, torch.Tensor) else outputs
+    ubatch_size = models.get_microbatch_size(outputs, verify=True)
+    rate = module.rate_constraint.item()
+    ubatch_time = torch.tensor(ubatch_size / rate if rate > 0
+                               else float("inf"))
+    ubatch_mbits = sum(t.numel() * t.element_size() for t in tensors)         * 8 / 1e6
+    src_bit = torch.tensor(tensors[0].element_size() * 8)
+    quant_bit = quantutil.constrain_max_bitwidth(
+        ubatch_time, torch.tensor(ubatch_mbits), torch.tensor(bandwidth),
+        src_bit)
+    module.quant_bit = max(torch.tensor(2), quant_bit) % src_bit
+    logger.info("Adaptive quantization (heuristic2): bitwidth=%d",
+                int(module.quant_bit))
+
+
+def set_quant_controller(module, outputs) -> None:
+    """EMSOFT'13-style integral controller splitting windows between two
+    bitwidths."""
+    try:
+        bw1 = module.bitwidth1.item()
+        bw2 = module.bitwidth2.item()
+        bw1_iters = module.bitwidth1_iters.item()
+    except AttributeError:
+        bw1 = bw2 = bw1_iters = 0
+    with monitoring.get_locked_context(MONITORING_KEY_SEND) as mctx:
+        tag = mctx.get_tag(key=MONITORING_KEY_SEND)
+        window_size = mctx.get_window_size(key=MONITORING_KEY_SEND)
+        heartrate = mctx.get_window_heartrate(key=MONITORING_KEY_SEND)
+    if tag > 0 and tag % window_size == 0:
+        with _MODULE_QUANT_CONTROLLERS_LOCK:
+            if module not in _MODULE_QUANT_CONTROLLERS:
+                bw_start = module.quant_bit.item() or max(BITWIDTHS)
+                _MODULE_QUANT_CONTROLLERS[module] =                     quantutil.AdaptiveBitwidthPerformanceController(
+                        0, BITWIDTHS, bw_start)
+        bw_ctlr = _MODULE_QUANT_CONTROLLERS[module]
+        bw_ctlr.reference = module.rate_constraint.item()
+        ubatch_size = models.get_microbatch_size(outputs, verify=True)
+        bw1, bw2, bw1_iters = bw_ctlr(heartrate * ubatch_size, window_size)
+        module.register_buffer("bitwidth1", torch.tensor(bw1),
+                               persistent=False)
+        module.register_buffer("bitwidth2", torch.tensor(bw2),
+                               persistent=False)
+        logger.info("Adaptive quantization (controller): bitwidth1=%d "
+                    "(iters=%d), bitwidth2=%d", bw1, bw1_iters, bw2)
+    bitwidth = bw1 if bw1_iters > 0 else bw2
+    module.quant_bit = torch.tensor(bitwidth % max(BITWIDTHS))
+    module.register_buffer("bitwidth1_iters",
+                           torch.tensor(max(0, bw1_iters - 1)),
+                           persistent=False)
+
+
+_ADAPTIVE_POLICIES = {
+    ADAPTIVE_QUANT_HEURISTIC: set_quant_bandwidth_heuristic,
+    ADAPTIVE_QUANT_HEURISTIC2: set_quant_bandwidth_heuristic2,
+    ADAPTIVE_QUANT_CONTROLLER: set_quant_controller,
+}
+
+
+def make_quant_send_transform(module, pad_words_per_tensor: List[int],
+                              max_bit: int):
+    """Encode stage outputs with the module's current (adaptive) quant_bit.
+
+    The static wire schema is sized for the stage's CONFIGURED bitwidth
+    (``max_bit``), so adaptive policies may only tighten compression:
+    bitwidths are clamped into [2, max_bit] (the reference could also turn
+    quantization fully off mid-run, which would change message sizes -
+    runtime.py:121-216; on xGMI the fixed ceiling-sized payload is not a
+    bottleneck)."""
+    policy = _ADAPTIVE_POLICIES.get(os.getenv(ENV_ADAPTIVE_QUANT, ""))
+
     def transform(outputs: models.ModuleShardData) -> List[torch.Tensor]:
+        if policy is not None:
+            try:
+                policy(module, outputs)
+            except Exception:  # noqa: BLE001
+                logger.exception("adaptive quant policy failed")
         monitoring.iteration_start(MONITORING_KEY_QUANT_ENCODE)
         if isinstance(outputs, torch.Tensor):
             outputs = (outputs,)
         quant_bit = int(module.quant_bit.item())
+        if quant_bit <= 0 or quant_bit > max_bit:
+            quant_bit = max_bit  # schema ceiling; see docstring
         wire: List[torch.Tensor] = []
         for tensor, pad_words in zip(outputs, pad_words_per_tensor):
             if quant_bit > 0:
@@ -326,6 +459,10 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
             module.register_buffer("quant_bit",
                                    torch.tensor(stage_quant[stage]),
                                    persistent=False)
+            send_constraint = float(os.getenv(ENV_SEND_CONSTRAINT, "0"))
+            module.register_buffer("rate_constraint",
+                                   torch.tensor(send_constraint),
+                                   persistent=False)
             module.register_forward_pre_hook(
                 lambda _m, _i: monitoring.iteration_start(MONITORING_KEY_MODEL))
 
@@ -354,7 +491,7 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
                 pad_words = [_pad_words(s, stage_quant[stage])
                              for s in out_specs]
                 stage_ctx.send_transform = make_quant_send_transform(
-                    module, pad_words)
+                    module, pad_words, stage_quant[stage])
             # inbound: dequantize unless we're stage 0
             if stage > 0 and stage_quant[stage - 1] > 0:
                 stage_ctx.recv_transform = make_quant_recv_transform(dtype)

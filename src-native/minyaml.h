@@ -256,10 +256,16 @@ class Parser {
             // indented deeper than the dash
             auto node = std::make_shared<Node>();
             node->kind = Node::Map;
+            auto next_is_seq = [&]() {
+                return pos_ < lines_.size() &&
+                       lines_[pos_].indent == indent + 2 &&
+                       (lines_[pos_].text.rfind("- ", 0) == 0 ||
+                        lines_[pos_].text == "-");
+            };
             if (!val.empty()) {
                 node->map.emplace_back(key, value_node(val, indent));
             } else if (pos_ < lines_.size() &&
-                       lines_[pos_].indent > indent + 2) {
+                       (lines_[pos_].indent > indent + 2 || next_is_seq())) {
                 node->map.emplace_back(key, parse_block(lines_[pos_].indent));
             }
             while (pos_ < lines_.size() && lines_[pos_].indent == indent + 2) {
@@ -269,7 +275,8 @@ class Parser {
                 if (!v2.empty()) {
                     node->map.emplace_back(k2, value_node(v2, indent + 2));
                 } else if (pos_ < lines_.size() &&
-                           lines_[pos_].indent > indent + 2) {
+                           (lines_[pos_].indent > indent + 2 ||
+                            next_is_seq())) {
                     node->map.emplace_back(k2,
                                            parse_block(lines_[pos_].indent));
                 }

@@ -1,0 +1,93 @@
+"""End-to-end runtime.py mode tests: RPC transport, adaptive quantization,
+auto-scheduling from YAML files (all 2-rank CPU/gloo subprocess runs)."""
+import multiprocessing as mp
+import os
+import subprocess
+import sys
+
+import pytest
+import yaml
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+MODEL = "facebook/deit-tiny-distilled-patch16-224"
+
+
+def _worker(rank, port, extra_args, env_extra, result_q, cwd):
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+               PYTHONPATH=REPO, **env_extra)
+    cmd = [sys.executable, os.path.join(REPO, "runtime.py"), str(rank), "2",
+           "-d", "cpu", "--port", str(port)] + extra_args
+    proc = subprocess.run(cmd, capture_output=True, timeout=300, env=env,
+                          cwd=cwd or REPO, check=False)
+    result_q.put((rank, proc.returncode, proc.stdout.decode(),
+                  proc.stderr.decode()[-1200:]))
+
+
+def _run_two(port, extra_args, env_extra=None, cwd=None):
+    ctx = mp.get_context("spawn")
+    result_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker,
+                         args=(r, port, extra_args, env_extra or {},
+                               result_q, cwd))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, rc, out, err = result_q.get(timeout=300)
+        outs[rank] = (rc, out, err)
+    for p in procs:
+        p.join(timeout=30)
+    return outs
+
+
+def test_runtime_rpc_two_rank():
+    outs = _run_two(29631, ["-c", "rpc", "-b", "4", "-u", "2", "-m", MODEL,
+                            "-pt", "1,24,25,48"])
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]
+
+
+def test_runtime_adaptive_quant_controller(tmp_path):
+    env = {"ADAPTIVE_QUANT": "CONTROLLER", "SEND_CONSTRAINT": "100",
+           "WINDOW_SIZE": "2"}
+    outs = _run_two(29633, ["-b", "8", "-u", "2", "-m", MODEL,
+                            "-pt", "1,24,25,48", "-q", "8,0"],
+                    env_extra=env, cwd=str(tmp_path))
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]
+    # the adaptive controller fired at the window boundary on rank 0
+    assert "Adaptive quantization (controller)" in outs[0][1], outs[0][1]
+
+
+def test_runtime_auto_sched_from_yaml(tmp_path):
+    layers = 48
+    model = {"layers": layers, "parameters_in": 150528,
+             "parameters_out": [2 * 198 * 192 if l % 4 in (1, 3)
+                                else 198 * 192 for l in range(1, layers)]
+             + [1000],
+             "mem_MB": [5.0] * layers}
+    dev_types = {"mi355x": {
+        "mem_MB": 100000, "bw_Mbps": 100000,
+        "model_profiles": {MODEL: [
+            {"dtype": "torch.float32", "batch_size": 2,
+             "time_s": [0.01] * layers}]}}}
+    devices_yml = {"mi355x": ["0", "1"]}
+    with open(tmp_path / "models.yml", "w") as f:
+        yaml.safe_dump({MODEL: model}, f)
+    with open(tmp_path / "device_types.yml", "w") as f:
+        yaml.safe_dump(dev_types, f)
+    with open(tmp_path / "devices.yml", "w") as f:
+        yaml.safe_dump(devices_yml, f)
+    extra = ["-b", "4", "-u", "2", "-m", MODEL,
+             "-sm", str(tmp_path / "models.yml"),
+             "-sdt", str(tmp_path / "device_types.yml"),
+             "-sd", str(tmp_path / "devices.yml")]
+    outs = _run_two(29635, extra, cwd=str(tmp_path))
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]
+    # monitoring CSVs were written by the run
+    assert (tmp_path / "shard.csv").exists()
