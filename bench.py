@@ -50,7 +50,8 @@ def run_rank(args, rank: int, world_size: int) -> dict:
 
     use_cuda = torch.cuda.is_available()
     if use_cuda:
-        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        # modulo lets N ranks share fewer GPUs (single-GPU RCCL validation)
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))             % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
         devices.initialize(f"cuda:{local_rank}")
     else:
