@@ -14,15 +14,16 @@
 // zero-padding LDS columns (dh=64 -> DHP=64; dh=80 ViT-Huge -> DHP=96).
 #include <torch/extension.h>
 
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
 
-constexpr int KVT = 64;   // kv positions per tile
 constexpr int QCH = 64;   // q rows per workgroup
 constexpr int PAD = 8;
 
-template <int DHP>
+template <int DHP, int KVT = 64>
 __global__ __launch_bounds__(256) void qkv_attention_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
         int B, int S, int H, int dh, float scale) {
@@ -201,10 +202,20 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     dim3 grid(B * H, (S + QCH - 1) / QCH);
     const auto* in = reinterpret_cast<const __bf16*>(x.data_ptr());
     auto* out = reinterpret_cast<__bf16*>(ctx.data_ptr());
+    static const int kvt = []() {
+        const char* e = std::getenv("PIPEEDGE_ATTN_KVT");
+        return e ? std::atoi(e) : 64;
+    }();
     if (dh <= 64) {
-        hipLaunchKernelGGL((qkv_attention_kernel<64>), grid, dim3(256), 0,
-                           pe_current_stream(), in, out, B, S, H, dh,
-                           (float)scale);
+        if (kvt >= 128 && S > 64) {
+            hipLaunchKernelGGL((qkv_attention_kernel<64, 128>), grid,
+                               dim3(256), 0, pe_current_stream(), in, out,
+                               B, S, H, dh, (float)scale);
+        } else {
+            hipLaunchKernelGGL((qkv_attention_kernel<64, 64>), grid,
+                               dim3(256), 0, pe_current_stream(), in, out,
+                               B, S, H, dh, (float)scale);
+        }
     } else if (dh <= 96) {
         hipLaunchKernelGGL((qkv_attention_kernel<96>), grid, dim3(256), 0,
                            pe_current_stream(), in, out, B, S, H, dh,

@@ -32,13 +32,22 @@ __device__ __forceinline__ void glds_b128(const void* gsrc, void* lds_dst) {
         (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
-// read a bf16x8 fragment from a [64][64] lane-linear tile with the
+// read a bf16x8 fragment from a lane-linear [rows][KB] tile with the
 // (row&7)<<4 byte XOR swizzle (matches the staging source permutation)
+template <int KB>  // K extent of the tile in elements
 __device__ __forceinline__ bf16x8 lds_read_swz(const __bf16* tile, int row,
                                                int kelem) {
-    const int byte = (row * 128 + kelem * 2) ^ ((row & 7) << 4);
+    const int byte = (row * (KB * 2) + kelem * 2) ^ ((row & 7) << 4);
     return *reinterpret_cast<const bf16x8*>(
         reinterpret_cast<const char*>(tile) + byte);
+}
+
+// bijective XCD-aware block remap (guide T1): contiguous grid chunks per XCD
+__device__ __forceinline__ int xcd_swizzle(int bid, int nblocks) {
+    const int nxcd = 8;
+    const int q = nblocks / nxcd, r = nblocks % nxcd;
+    const int xcd = bid % nxcd, i = bid / nxcd;
+    return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
 }
 
 struct EpilogueArgs {
@@ -79,41 +88,52 @@ __device__ __forceinline__ void gemm_epilogue(
 // each wave an MIxNI grid of 16x16 fragments. <2,2> = 64x64 (small-N
 // shapes need the block count), <4,4> = 128x128 (fc1/qkv class).
 // ---------------------------------------------------------------------------
-template <int MI, int NI>
+template <int MI, int NI, int KT = 1, bool XCD = false>
 __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
         __bf16* __restrict__ C, int M, int N, int K, int act) {
     constexpr int TBM = 32 * MI;
     constexpr int TBN = 32 * NI;
-    // ONE shared object (guide section 5 trap 4a): [buf][TBM+TBN rows][64]
-    __shared__ __bf16 smem[2][TBM + TBN][BK];
+    constexpr int KB = BK * KT;          // K extent per staged tile
+    // ONE shared object (guide section 5 trap 4a): [buf][TBM+TBN rows][KB]
+    __shared__ __bf16 smem[2][TBM + TBN][KB];
 
-    const int bm = blockIdx.x * TBM;
-    const int bn = blockIdx.y * TBN;
+    int bx = blockIdx.x, by = blockIdx.y;
+    if (XCD) {
+        const int nb = gridDim.x * gridDim.y;
+        const int id = xcd_swizzle(by * gridDim.x + bx, nb);
+        bx = id % gridDim.x;
+        by = id / gridDim.x;
+    }
+    const int bm = bx * TBM;
+    const int bn = by * TBN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
     const int wm = (wid >> 1) * (16 * MI);
     const int wn = (wid & 1) * (16 * NI);
 
-    // lane's role in staging: row-in-piece and swizzled source column
-    const int srow = lane >> 3;              // 0..7 within the 8-row piece
-    const int csrc = (lane & 7) ^ srow;      // XOR-swizzled 16-B column
+    // staging: 1-KiB glds pieces of R=8/KT rows x (8*KT) 16-B columns each;
+    // source column XOR-swizzled to match the read-side (row&7)<<4 swizzle
+    constexpr int RPP = 8 / KT;              // rows per piece
+    const int srow = lane / (8 * KT);        // row within piece
+    const int cdst = lane % (8 * KT);
+    const int csrc = cdst ^ (srow & 7);
 
     auto stage = [&](int buf, int k0) {
         #pragma unroll
-        for (int i = 0; i < TBM / 32; ++i) {
-            const int p = wid * (TBM / 32) + i;  // rows p*8..p*8+7 of A
-            const long arow = min(bm + p * 8 + srow, M - 1);
-            glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * 8][0]);
+        for (int i = 0; i < TBM / (32 / KT); ++i) {
+            const int p = wid * (TBM / (32 / KT)) + i;
+            const long arow = min(bm + p * RPP + srow, M - 1);
+            glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * RPP][0]);
         }
         #pragma unroll
-        for (int i = 0; i < TBN / 32; ++i) {
-            const int p = wid * (TBN / 32) + i;  // rows p*8..p*8+7 of B
-            const long brow = min(bn + p * 8 + srow, N - 1);
+        for (int i = 0; i < TBN / (32 / KT); ++i) {
+            const int p = wid * (TBN / (32 / KT)) + i;
+            const long brow = min(bn + p * RPP + srow, N - 1);
             glds_b128(B + brow * K + k0 + csrc * 8,
-                      &smem[buf][TBM + p * 8][0]);
+                      &smem[buf][TBM + p * RPP][0]);
         }
     };
 
@@ -121,22 +141,22 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
     stage(0, 0);
     __syncthreads();  // barrier waits vmcnt(0) while glds in flight
 
-    const int n_tiles = K / BK;
+    const int n_tiles = K / KB;
     for (int t = 0; t < n_tiles; ++t) {
         const int buf = t & 1;
-        if (t + 1 < n_tiles) stage(buf ^ 1, (t + 1) * BK);
+        if (t + 1 < n_tiles) stage(buf ^ 1, (t + 1) * KB);
         #pragma unroll
-        for (int kk = 0; kk < BK; kk += 32) {
+        for (int kk = 0; kk < KB; kk += 32) {
             const int kgrp = kk + (lane >> 4) * 8;
             bf16x8 af[MI], bfr[NI];
             #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
-                af[mi] = lds_read_swz(&smem[buf][0][0],
-                                      wm + mi * 16 + (lane & 15), kgrp);
+                af[mi] = lds_read_swz<KB>(&smem[buf][0][0],
+                                          wm + mi * 16 + (lane & 15), kgrp);
             #pragma unroll
             for (int ni = 0; ni < NI; ++ni)
-                bfr[ni] = lds_read_swz(&smem[buf][TBM][0],
-                                       wn + ni * 16 + (lane & 15), kgrp);
+                bfr[ni] = lds_read_swz<KB>(&smem[buf][TBM][0],
+                                           wn + ni * 16 + (lane & 15), kgrp);
             #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
                 #pragma unroll
@@ -266,7 +286,25 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
             // only for wide-N long-M (fc1 class)
             t = (N >= 3072 && M >= 1024) ? 2 : 1;
         }
-        if (t == 3) {
+        if (t == 4 && K % (2 * BK) != 0) t = 1;
+        if (t == 6 && K % (2 * BK) != 0) t = 5;
+        if (t == 4 || t == 5 || t == 6) {
+            dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+            if (t == 4)
+                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2>), grid,
+                                   dim3(256), 0, pe_current_stream(), ap, bp,
+                                   bias_p, res_p, cp, M, N, K, (int)act);
+            else if (t == 5)
+                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 1, true>),
+                                   grid, dim3(256), 0, pe_current_stream(),
+                                   ap, bp, bias_p, res_p, cp, M, N, K,
+                                   (int)act);
+            else
+                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2, true>),
+                                   grid, dim3(256), 0, pe_current_stream(),
+                                   ap, bp, bias_p, res_p, cp, M, N, K,
+                                   (int)act);
+        } else if (t == 3) {
             dim3 grid((M + 127) / 128, (N + 127) / 128);
             hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 4>), grid,
                                dim3(256), 0, pe_current_stream(), ap, bp,
