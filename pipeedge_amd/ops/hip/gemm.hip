@@ -119,19 +119,24 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
     constexpr int RPP = 8 / KT;              // rows per piece
     const int srow = lane / (8 * KT);        // row within piece
     const int cdst = lane % (8 * KT);
-    const int csrc = cdst ^ (srow & 7);
 
+    // the XOR swizzle keys off the GLOBAL tile row mod 8, which spans
+    // multiple pieces when RPP < 8
     auto stage = [&](int buf, int k0) {
         #pragma unroll
         for (int i = 0; i < TBM / (32 / KT); ++i) {
             const int p = wid * (TBM / (32 / KT)) + i;
-            const long arow = min(bm + p * RPP + srow, M - 1);
+            const int trow = p * RPP + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long arow = min(bm + trow, M - 1);
             glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * RPP][0]);
         }
         #pragma unroll
         for (int i = 0; i < TBN / (32 / KT); ++i) {
             const int p = wid * (TBN / (32 / KT)) + i;
-            const long brow = min(bn + p * RPP + srow, N - 1);
+            const int trow = p * RPP + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long brow = min(bn + trow, N - 1);
             glds_b128(B + brow * K + k0 + csrc * 8,
                       &smem[buf][TBM + p * RPP][0]);
         }
@@ -281,10 +286,12 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
         // the `tile` arg (0 = auto heuristic)
         int t = (int)tile;
         if (t == 0) {
-            // from the on-HW sweep (tools/gemm_sweep.py): 64x64 wins at
-            // transformer-inference shapes (grid occupancy bound); 128x64
-            // only for wide-N long-M (fc1 class)
-            t = (N >= 3072 && M >= 1024) ? 2 : 1;
+            // from the on-HW sweeps (tools/gemm_sweep.py,
+            // tools/hotpath_sweep.py): XCD-swizzled 64x64 default; BK=128
+            // for long-K (fc2 class, +27%); 128x64 for wide-N (fc1 class)
+            if (K >= 2048 && K % (2 * BK) == 0) t = 6;
+            else if (N >= 3072 && M >= 1024) t = 2;
+            else t = 5;
         }
         if (t == 4 && K % (2 * BK) != 0) t = 1;
         if (t == 6 && K % (2 * BK) != 0) t = 5;
