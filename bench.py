@@ -75,10 +75,12 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     module = model_cfg.module_shard_factory(
         model_name, args.model_file, stage_layers[stage][0],
         stage_layers[stage][1], stage, dtype=dtype)
+    from pipeedge_amd.utils.graph import maybe_graph
+    work_cb = maybe_graph(module)
 
     results = []
     stage_ctx = model_cfg.dist_p2p_pipeline_stage_factory(
-        stage_ranks, data_rank, rank, stage, module, results.append,
+        stage_ranks, data_rank, rank, stage, work_cb, results.append,
         model_name=model_name, ubatch_size=args.ubatch_size,
         stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype)
 

@@ -463,19 +463,25 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
             module.register_buffer("rate_constraint",
                                    torch.tensor(send_constraint),
                                    persistent=False)
-            module.register_forward_pre_hook(
-                lambda _m, _i: monitoring.iteration_start(MONITORING_KEY_MODEL))
 
-            def _fwd_monitor(mod, _inputs, outputs):
+
+        # hipGraph-replayed forward on GPU; shard monitoring wraps the
+        # callable (module forward hooks would not fire during graph replay)
+        from pipeedge_amd.utils.graph import maybe_graph
+        work_cb = None
+        if module is not None:
+            inner_cb = maybe_graph(module)
+            n_layers = stage_layers[stage][1] - stage_layers[stage][0] + 1
+
+            def work_cb(data, _cb=inner_cb, _n_layers=n_layers):
+                monitoring.iteration_start(MONITORING_KEY_MODEL)
+                outputs = _cb(data)
                 n_items = models.get_microbatch_size(outputs, verify=True)
-                n_layers = mod.shard_config.layer_end - \
-                    mod.shard_config.layer_start + 1
                 monitoring.iteration(MONITORING_KEY_MODEL, work=n_items,
-                                     accuracy=n_layers)
-            module.register_forward_hook(_fwd_monitor)
-
+                                     accuracy=_n_layers)
+                return outputs
         stage_ctx = model_cfg.dist_p2p_pipeline_stage_factory(
-            stage_ranks, data_rank, rank, stage, module, handle_results,
+            stage_ranks, data_rank, rank, stage, work_cb, handle_results,
             model_name=model_name, ubatch_size=ubatch_size,
             stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype)
 
