@@ -132,26 +132,80 @@ class ViTLayerShard(ModuleShard):
                 kind = "trunc"
             init_param(p, f"{prefix}/block{self.block_id}/{name}", kind)
 
+    def _ln_folds(self):
+        """Lazy folded-LN GEMM operands (GPU path; see ops.make_ln_fold)."""
+        if getattr(self, "_fold_cache", None) is None:
+            cache = {}
+            with torch.no_grad():
+                if self.has_layer(0):
+                    cache["qkv"] = ops.make_ln_fold(
+                        self.qkv_weight, self.ln1_weight, self.ln1_bias)
+                if self.has_layer(2):
+                    cache["fc1"] = ops.make_ln_fold(
+                        self.fc1_weight, self.ln2_weight, self.ln2_bias)
+            self._fold_cache = cache
+        return self._fold_cache
+
+    @staticmethod
+    def _stats_buf(t):
+        return torch.zeros((t.shape[0] * t.shape[1], 2), dtype=torch.float32,
+                           device=t.device)
+
     @torch.no_grad()
-    def forward(self, data):
+    def forward(self, data, stats=None):
+        """Compute the shard's sublayers.
+
+        ``stats``: per-row (sum, sumsq) of the incoming hidden state when the
+        producing GEMM emitted them (GPU folded-LN path); returns
+        (data, stats) when called via forward_with_stats.
+        """
+        data, _ = self.forward_with_stats(data, stats)
+        return data
+
+    @torch.no_grad()
+    def forward_with_stats(self, data, stats=None):
         cfg = self.config
         eps = cfg.layer_norm_eps
         if self.has_layer(0):
-            normed = ops.layernorm(data, self.ln1_weight, self.ln1_bias, eps)
-            qkv = ops.linear(normed, self.qkv_weight, self.qkv_bias)
+            if ops.fused_ln_active(data):
+                qkv = ops.ln_linear(data, self.ln1_weight, self.ln1_bias,
+                                    eps, self.qkv_weight, self.qkv_bias,
+                                    stats=stats,
+                                    fused=self._ln_folds()["qkv"])
+            else:
+                normed = ops.layernorm(data, self.ln1_weight, self.ln1_bias,
+                                       eps)
+                qkv = ops.linear(normed, self.qkv_weight, self.qkv_bias)
             ctx = ops.qkv_attention(qkv, cfg.num_attention_heads, self.scale)
             data = (ctx, data)
+            stats = None
         if self.has_layer(1):
+            stats = self._stats_buf(data[0])                 if ops.fused_ln_active(data[0]) else None
             data = ops.linear(data[0], self.out_weight, self.out_bias,
-                              residual=data[1])
+                              residual=data[1], stats_out=stats)
         if self.has_layer(2):
-            normed = ops.layernorm(data, self.ln2_weight, self.ln2_bias, eps)
-            mid = ops.linear(normed, self.fc1_weight, self.fc1_bias, act="gelu")
+            if ops.fused_ln_active(data):
+                mid = ops.ln_linear(data, self.ln2_weight, self.ln2_bias,
+                                    eps, self.fc1_weight, self.fc1_bias,
+                                    act="gelu", stats=stats,
+                                    fused=self._ln_folds()["fc1"])
+            else:
+                normed = ops.layernorm(data, self.ln2_weight, self.ln2_bias,
+                                       eps)
+                mid = ops.linear(normed, self.fc1_weight, self.fc1_bias,
+                                 act="gelu")
             data = (mid, data)
+            stats = None
         if self.has_layer(3):
+            stats = self._stats_buf(data[0].reshape(
+                data[1].shape[0], data[1].shape[1], -1)[:, :, :1])                 if ops.fused_ln_active(data[0]) else None
+            if stats is not None:
+                stats = torch.zeros(
+                    (data[1].shape[0] * data[1].shape[1], 2),
+                    dtype=torch.float32, device=data[1].device)
             data = ops.linear(data[0], self.fc2_weight, self.fc2_bias,
-                              residual=data[1])
-        return data
+                              residual=data[1], stats_out=stats)
+        return data, stats
 
 
 class ViTModelShard(ModuleShard):
@@ -240,8 +294,9 @@ class ViTModelShard(ModuleShard):
     def forward(self, data):
         if self.shard_config.is_first:
             data = self.embeddings(data)
+        stats = None  # per-row hidden stats threaded between folded-LN GEMMs
         for layer in self.layers:
-            data = layer(data)
+            data, stats = layer.forward_with_stats(data, stats)
         if self.shard_config.is_last:
             data = ops.layernorm(data, self.layernorm_weight, self.layernorm_bias,
                                  self.config.layer_norm_eps)

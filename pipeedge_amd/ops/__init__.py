@@ -77,10 +77,63 @@ _ACT_CODE = {None: 0, "gelu": 1, "tanh": 2}
 
 
 def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
-           act: Optional[str] = None, residual: Optional[Tensor] = None) -> Tensor:
+           act: Optional[str] = None, residual: Optional[Tensor] = None,
+           stats_out: Optional[Tensor] = None) -> Tensor:
+    """GEMM + fused epilogue. ``stats_out`` (GPU path): zeroed f32 [rows, 2]
+    buffer that receives per-row (sum, sumsq) of the OUTPUT - seeds the next
+    folded-LN GEMM (see ln_linear)."""
     if _use_hip(x):
-        return _hip.linear(x, weight_nk, bias, _ACT_CODE[act], residual)
+        return _hip.linear(x, weight_nk, bias, _ACT_CODE[act], residual,
+                           0, None, None, None, 1e-12, stats_out)
     return torch_ref.linear(x, weight_nk, bias, act, residual)
+
+
+def ln_stats(x: Tensor) -> Tensor:
+    """Per-row (sum, sumsq) over the last dim, f32 [rows, 2]."""
+    if _use_hip(x):
+        return _hip.ln_stats(x.reshape(-1, x.shape[-1]))
+    xf = x.reshape(-1, x.shape[-1]).float()
+    return torch.stack((xf.sum(-1), (xf * xf).sum(-1)), dim=-1)
+
+
+def make_ln_fold(weight_nk: Tensor, ln_weight: Tensor, ln_bias: Tensor):
+    """Precompute the folded-LN GEMM operands (GPU):
+    W' = W * ln_w (per-K column), s[n] = sum_k W'[n,k], c[n] = W @ ln_b.
+
+    Because LayerNorm is affine per row,
+      LN(x) @ W^T = r*(x @ W'^T) - r*mu*s + c
+    with per-row mean mu and inv-std r - so the GEMM runs on RAW x and the
+    LN collapses into the epilogue (no standalone LN kernel, no extra
+    activation pass)."""
+    w_p = (weight_nk.float() * ln_weight.float()).to(weight_nk.dtype)         .contiguous()
+    s = w_p.float().sum(dim=1).contiguous()
+    c = (weight_nk.float() @ ln_bias.float()).contiguous()
+    return (w_p, s, c)
+
+
+def ln_linear(x: Tensor, ln_weight: Tensor, ln_bias: Tensor, eps: float,
+              weight_nk: Tensor, bias: Optional[Tensor] = None,
+              act: Optional[str] = None, residual: Optional[Tensor] = None,
+              stats: Optional[Tensor] = None, fused=None,
+              stats_out: Optional[Tensor] = None) -> Tensor:
+    """LayerNorm -> Linear; folded into ONE GEMM on GPU (see make_ln_fold).
+
+    ``stats``: per-row (sum, sumsq) of x if already known (emitted by the
+    producing GEMM); computed by a ln_stats kernel otherwise."""
+    if _use_hip(x):
+        assert fused is not None, "ln_linear GPU path needs make_ln_fold(...)"
+        w_p, s, c = fused
+        if stats is None:
+            stats = _hip.ln_stats(x.reshape(-1, x.shape[-1]))
+        return _hip.linear(x, w_p, bias, _ACT_CODE[act], residual, 0,
+                           stats, s, c, eps, stats_out)
+    y = torch_ref.layernorm(x, ln_weight, ln_bias, eps)
+    return torch_ref.linear(y, weight_nk, bias, act, residual)
+
+
+def fused_ln_active(x: Tensor) -> bool:
+    """Whether the folded-LN GEMM path will be used for this tensor."""
+    return x.is_cuda and not force_torch()
 
 
 def attention_core(q: Tensor, k: Tensor, v: Tensor, scale: Optional[float] = None,

@@ -7,7 +7,13 @@
 torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                         c10::optional<torch::Tensor> bias, int64_t act,
                         c10::optional<torch::Tensor> residual2d,
-                        int64_t tile = 0);
+                        int64_t tile = 0,
+                        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
+                        c10::optional<torch::Tensor> ln_s = c10::nullopt,
+                        c10::optional<torch::Tensor> ln_c = c10::nullopt,
+                        double ln_eps = 1e-12,
+                        c10::optional<torch::Tensor> stats_out = c10::nullopt);
+torch::Tensor ln_stats_bf16(torch::Tensor x);
 torch::Tensor debug_mfma_tile(torch::Tensor a, torch::Tensor b);
 torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor weight,
                              torch::Tensor bias, double eps);
@@ -24,14 +30,20 @@ namespace {
 torch::Tensor linear_op(torch::Tensor x, torch::Tensor w,
                         c10::optional<torch::Tensor> bias, int64_t act,
                         c10::optional<torch::Tensor> residual,
-                        int64_t tile = 0) {
+                        int64_t tile = 0,
+                        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
+                        c10::optional<torch::Tensor> ln_s = c10::nullopt,
+                        c10::optional<torch::Tensor> ln_c = c10::nullopt,
+                        double ln_eps = 1e-12,
+                        c10::optional<torch::Tensor> stats_out = c10::nullopt) {
     auto sizes = x.sizes().vec();
     const long K = sizes.back();
     auto x2d = x.reshape({-1, K});
     c10::optional<torch::Tensor> res2d;
     if (residual.has_value())
         res2d = residual->reshape({-1, w.size(0)});  // w is [N, K]
-    auto out = gemm_bf16(x2d, w, bias, act, res2d, tile);
+    auto out = gemm_bf16(x2d, w, bias, act, res2d, tile, ln_stats, ln_s,
+                         ln_c, ln_eps, stats_out);
     sizes.back() = w.size(0);
     return out.reshape(sizes);
 }
@@ -50,7 +62,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("linear", &linear_op, "fused GEMM+bias+act+residual (bf16 MFMA)",
           py::arg("x"), py::arg("weight"), py::arg("bias") = py::none(),
           py::arg("act") = 0, py::arg("residual") = py::none(),
-          py::arg("tile") = 0);
+          py::arg("tile") = 0, py::arg("ln_stats") = py::none(),
+          py::arg("ln_s") = py::none(), py::arg("ln_c") = py::none(),
+          py::arg("ln_eps") = 1e-12, py::arg("stats_out") = py::none());
+    m.def("ln_stats", &ln_stats_bf16, py::arg("x"));
     m.def("layernorm", &layernorm_bf16, py::arg("x"), py::arg("weight"),
           py::arg("bias"), py::arg("eps") = 1e-6);
     m.def("add_layernorm", &add_layernorm_bf16, py::arg("x"),

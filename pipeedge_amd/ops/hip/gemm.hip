@@ -55,12 +55,33 @@ struct EpilogueArgs {
     const __bf16* residual;
     __bf16* C;
     int M, N, act;
+    // folded-LN input adjustment (see ops/__init__.py ln_linear): the GEMM
+    // ran on RAW x against W' = W*ln_w; the epilogue applies
+    //   out = r*acc - r*mu*ln_s[col] + ln_c[col]
+    // with per-row mu/r derived from ln_stats[row] = (sum, sumsq) over K.
+    const float* ln_stats;   // [M][2] or nullptr
+    const float* ln_s;       // [N] row-sums of W'
+    const float* ln_c;       // [N] ln_b @ W^T
+    float ln_eps;
+    int ln_k;                // LN normalization length (= GEMM K)
+    // output-stats emission: per-row (sum, sumsq) of C accumulated with
+    // atomics into stats_out[M][2] (zeroed by the host before launch)
+    float* stats_out;
 };
 
 template <int MI, int NI>
 __device__ __forceinline__ void gemm_epilogue(
         const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
         int wm, int wn, int lane) {
+    float row_sum[MI][4];
+    float row_sumsq[MI][4];
+    if (ep.stats_out != nullptr) {
+        #pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                row_sum[mi][r] = row_sumsq[mi][r] = 0.0f;
+    }
     #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
         #pragma unroll
@@ -68,16 +89,48 @@ __device__ __forceinline__ void gemm_epilogue(
             const int col = bn + wn + ni * 16 + (lane & 15);
             if (col >= ep.N) continue;
             const float bv = ep.bias != nullptr ? bf2f(ep.bias[col]) : 0.0f;
+            const float sv = ep.ln_stats != nullptr ? ep.ln_s[col] : 0.0f;
+            const float cv = ep.ln_stats != nullptr ? ep.ln_c[col] : 0.0f;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= ep.M) continue;
-                float v = acc[mi][ni][r] + bv;
+                float v = acc[mi][ni][r];
+                if (ep.ln_stats != nullptr) {
+                    const float s1 = ep.ln_stats[row * 2];
+                    const float s2 = ep.ln_stats[row * 2 + 1];
+                    const float mu = s1 / ep.ln_k;
+                    const float rstd = rsqrtf(s2 / ep.ln_k - mu * mu +
+                                              ep.ln_eps);
+                    v = rstd * v - rstd * mu * sv + cv;
+                }
+                v += bv;
                 if (ep.act == ACT_GELU) v = gelu_f(v);
                 else if (ep.act == ACT_TANH) v = tanhf(v);
                 if (ep.residual != nullptr)
                     v += bf2f(ep.residual[(long)row * ep.N + col]);
                 ep.C[(long)row * ep.N + col] = f2bf(v);
+                if (ep.stats_out != nullptr) {
+                    row_sum[mi][r] += v;
+                    row_sumsq[mi][r] += v * v;
+                }
+            }
+        }
+    }
+    if (ep.stats_out != nullptr) {
+        // reduce across the 16 lanes sharing each C-fragment row, then one
+        // atomic pair per (wave, mi, r)
+        #pragma unroll
+        for (int mi = 0; mi < MI; ++mi) {
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float s1 = group16_reduce(row_sum[mi][r], SumOp());
+                float s2 = group16_reduce(row_sumsq[mi][r], SumOp());
+                const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
+                if ((lane & 15) == 0 && row < ep.M) {
+                    atomicAdd(&ep.stats_out[row * 2], s1);
+                    atomicAdd(&ep.stats_out[row * 2 + 1], s2);
+                }
             }
         }
     }
@@ -91,8 +144,7 @@ __device__ __forceinline__ void gemm_epilogue(
 template <int MI, int NI, int KT = 1, bool XCD = false>
 __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
-        const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
-        __bf16* __restrict__ C, int M, int N, int K, int act) {
+        __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
     constexpr int TBM = 32 * MI;
     constexpr int TBN = 32 * NI;
     constexpr int KB = BK * KT;          // K extent per staged tile
@@ -172,8 +224,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         __syncthreads();
     }
 
-    gemm_epilogue<MI, NI>(acc, {bias, residual, C, M, N, act}, bm, bn, wm,
-                          wn, lane);
+    gemm_epilogue<MI, NI>(acc, ep, bm, bn, wm, wn, lane);
 }
 
 // ---------------------------------------------------------------------------
@@ -181,8 +232,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
-        const __bf16* __restrict__ bias, const __bf16* __restrict__ residual,
-        __bf16* __restrict__ C, int M, int N, int K, int act) {
+        __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
     __shared__ __bf16 As[BM][BK + PAD];
     __shared__ __bf16 Bs[BN][BK + PAD];
 
@@ -244,8 +294,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         __syncthreads();
     }
 
-    gemm_epilogue<2, 2>(acc, {bias, residual, C, M, N, act}, bm, bn, wm,
-                        wn, lane);
+    gemm_epilogue<2, 2>(acc, ep, bm, bn, wm, wn, lane);
 }
 
 }  // namespace
@@ -253,7 +302,11 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
 torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                         c10::optional<torch::Tensor> bias, int64_t act,
                         c10::optional<torch::Tensor> residual2d,
-                        int64_t tile) {
+                        int64_t tile,
+                        c10::optional<torch::Tensor> ln_stats,
+                        c10::optional<torch::Tensor> ln_s,
+                        c10::optional<torch::Tensor> ln_c, double ln_eps,
+                        c10::optional<torch::Tensor> stats_out) {
     TORCH_CHECK(a2d.is_cuda() && a2d.dtype() == torch::kBFloat16,
                 "gemm: A must be bf16 on GPU");
     TORCH_CHECK(b.dtype() == torch::kBFloat16, "gemm: B must be bf16");
@@ -281,6 +334,24 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
     const auto* ap = reinterpret_cast<const __bf16*>(a_c.data_ptr());
     const auto* bp = reinterpret_cast<const __bf16*>(b_c.data_ptr());
     auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
+
+    EpilogueArgs ep{bias_p, res_p, cp, M, N, (int)act,
+                    nullptr, nullptr, nullptr, (float)ln_eps, K, nullptr};
+    if (ln_stats.has_value()) {
+        TORCH_CHECK(ln_s.has_value() && ln_c.has_value(),
+                    "gemm: ln_stats requires ln_s and ln_c");
+        TORCH_CHECK(ln_stats->dtype() == torch::kFloat32 &&
+                    ln_stats->numel() == 2 * M);
+        TORCH_CHECK(ln_s->numel() == N && ln_c->numel() == N);
+        ep.ln_stats = ln_stats->data_ptr<float>();
+        ep.ln_s = ln_s->data_ptr<float>();
+        ep.ln_c = ln_c->data_ptr<float>();
+    }
+    if (stats_out.has_value()) {
+        TORCH_CHECK(stats_out->dtype() == torch::kFloat32 &&
+                    stats_out->numel() == 2 * M);
+        ep.stats_out = stats_out->data_ptr<float>();
+    }
     if (K % BK == 0) {
         // tile choice: empirically swept (tests/gpu sweep); override via
         // the `tile` arg (0 = auto heuristic)
@@ -300,38 +371,36 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
             if (t == 4)
                 hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2>), grid,
                                    dim3(256), 0, pe_current_stream(), ap, bp,
-                                   bias_p, res_p, cp, M, N, K, (int)act);
+                                   cp, M, N, K, ep);
             else if (t == 5)
                 hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 1, true>),
                                    grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, bias_p, res_p, cp, M, N, K,
-                                   (int)act);
+                                   ap, bp, cp, M, N, K, ep);
             else
                 hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2, true>),
                                    grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, bias_p, res_p, cp, M, N, K,
-                                   (int)act);
+                                   ap, bp, cp, M, N, K, ep);
         } else if (t == 3) {
             dim3 grid((M + 127) / 128, (N + 127) / 128);
             hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 4>), grid,
                                dim3(256), 0, pe_current_stream(), ap, bp,
-                               bias_p, res_p, cp, M, N, K, (int)act);
+                               cp, M, N, K, ep);
         } else if (t == 2) {
             dim3 grid((M + 127) / 128, (N + 63) / 64);
             hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 2>), grid,
                                dim3(256), 0, pe_current_stream(), ap, bp,
-                               bias_p, res_p, cp, M, N, K, (int)act);
+                               cp, M, N, K, ep);
         } else {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2>), grid,
                                dim3(256), 0, pe_current_stream(), ap, bp,
-                               bias_p, res_p, cp, M, N, K, (int)act);
+                               cp, M, N, K, ep);
         }
     } else {
         dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
         hipLaunchKernelGGL(gemm_bf16_generic_kernel, grid, dim3(256), 0,
-                           pe_current_stream(), ap, bp, bias_p, res_p, cp,
-                           M, N, K, (int)act);
+                           pe_current_stream(), ap, bp, cp,
+                           M, N, K, ep);
     }
     HIP_CHECK_LAST();
     return out;

@@ -43,3 +43,15 @@ gb = 1576*768*2*2/1e9
 print(f"LN: {us:.2f} us = {gb/(us*1e-6):.2f} GB/s eff")
 us = bench(lambda: hip.add_layernorm(x,x,w,bb,1e-6))
 print(f"addLN: {us:.2f} us")
+
+print("== LN split A/B (env PIPEEDGE_LN_SPLIT)")
+import os
+print("split =", os.environ.get("PIPEEDGE_LN_SPLIT", "default-on"))
+for rows, d in [(1576,768),(1576,1024),(4096,768)]:
+    x = torch.randn(rows,d).bfloat16().cuda()
+    w = torch.ones(d).bfloat16().cuda(); bb = torch.zeros(d).bfloat16().cuda()
+    ref = torch.nn.functional.layer_norm(x.float(), (d,), w.float(), bb.float(), 1e-6)
+    got = hip.layernorm(x,w,bb,1e-6)
+    assert (got.float()-ref).abs().max().item() < 0.05, (rows,d)
+    us = bench(lambda: hip.layernorm(x,w,bb,1e-6))
+    print(f"LN {rows}x{d}: {us:.2f} us  ({rows*d*2*2/1e3/us:.0f} GB/s)")
