@@ -146,11 +146,6 @@ class ViTLayerShard(ModuleShard):
             self._fold_cache = cache
         return self._fold_cache
 
-    @staticmethod
-    def _stats_buf(t):
-        return torch.zeros((t.shape[0] * t.shape[1], 2), dtype=torch.float32,
-                           device=t.device)
-
     @torch.no_grad()
     def forward(self, data, stats=None):
         """Compute the shard's sublayers.
@@ -180,9 +175,14 @@ class ViTLayerShard(ModuleShard):
             data = (ctx, data)
             stats = None
         if self.has_layer(1):
-            stats = self._stats_buf(data[0])                 if ops.fused_ln_active(data[0]) else None
-            data = ops.linear(data[0], self.out_weight, self.out_bias,
-                              residual=data[1], stats_out=stats)
+            if ops.fused_ln_active(data[0]):
+                data, stats = ops.linear(data[0], self.out_weight,
+                                         self.out_bias, residual=data[1],
+                                         want_stats=True)
+            else:
+                data = ops.linear(data[0], self.out_weight, self.out_bias,
+                                  residual=data[1])
+                stats = None
         if self.has_layer(2):
             if ops.fused_ln_active(data):
                 mid = ops.ln_linear(data, self.ln2_weight, self.ln2_bias,
@@ -197,14 +197,14 @@ class ViTLayerShard(ModuleShard):
             data = (mid, data)
             stats = None
         if self.has_layer(3):
-            stats = self._stats_buf(data[0].reshape(
-                data[1].shape[0], data[1].shape[1], -1)[:, :, :1])                 if ops.fused_ln_active(data[0]) else None
-            if stats is not None:
-                stats = torch.zeros(
-                    (data[1].shape[0] * data[1].shape[1], 2),
-                    dtype=torch.float32, device=data[1].device)
-            data = ops.linear(data[0], self.fc2_weight, self.fc2_bias,
-                              residual=data[1], stats_out=stats)
+            if ops.fused_ln_active(data[0]):
+                data, stats = ops.linear(data[0], self.fc2_weight,
+                                         self.fc2_bias, residual=data[1],
+                                         want_stats=True)
+            else:
+                data = ops.linear(data[0], self.fc2_weight, self.fc2_bias,
+                                  residual=data[1])
+                stats = None
         return data, stats
 
 

@@ -78,22 +78,26 @@ _ACT_CODE = {None: 0, "gelu": 1, "tanh": 2}
 
 def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
            act: Optional[str] = None, residual: Optional[Tensor] = None,
-           stats_out: Optional[Tensor] = None) -> Tensor:
-    """GEMM + fused epilogue. ``stats_out`` (GPU path): zeroed f32 [rows, 2]
-    buffer that receives per-row (sum, sumsq) of the OUTPUT - seeds the next
+           want_stats: bool = False):
+    """GEMM + fused epilogue. With ``want_stats`` (GPU path) also returns
+    per-row partial (sum, sumsq) of the OUTPUT, f32 [parts, rows, 2] -
+    written without atomics by each column block and summed by the next
     folded-LN GEMM (see ln_linear)."""
     if _use_hip(x):
-        return _hip.linear(x, weight_nk, bias, _ACT_CODE[act], residual,
-                           0, None, None, None, 1e-12, stats_out)
-    return torch_ref.linear(x, weight_nk, bias, act, residual)
+        out, stats = _hip.linear(x, weight_nk, bias, _ACT_CODE[act],
+                                 residual, 0, None, None, None, 1e-12,
+                                 want_stats)
+        return (out, stats) if want_stats else out
+    out = torch_ref.linear(x, weight_nk, bias, act, residual)
+    return (out, None) if want_stats else out
 
 
 def ln_stats(x: Tensor) -> Tensor:
-    """Per-row (sum, sumsq) over the last dim, f32 [rows, 2]."""
+    """Per-row (sum, sumsq) over the last dim, f32 [1, rows, 2]."""
     if _use_hip(x):
         return _hip.ln_stats(x.reshape(-1, x.shape[-1]))
     xf = x.reshape(-1, x.shape[-1]).float()
-    return torch.stack((xf.sum(-1), (xf * xf).sum(-1)), dim=-1)
+    return torch.stack((xf.sum(-1), (xf * xf).sum(-1)), dim=-1).unsqueeze(0)
 
 
 def make_ln_fold(weight_nk: Tensor, ln_weight: Tensor, ln_bias: Tensor):
@@ -114,8 +118,7 @@ def make_ln_fold(weight_nk: Tensor, ln_weight: Tensor, ln_bias: Tensor):
 def ln_linear(x: Tensor, ln_weight: Tensor, ln_bias: Tensor, eps: float,
               weight_nk: Tensor, bias: Optional[Tensor] = None,
               act: Optional[str] = None, residual: Optional[Tensor] = None,
-              stats: Optional[Tensor] = None, fused=None,
-              stats_out: Optional[Tensor] = None) -> Tensor:
+              stats: Optional[Tensor] = None, fused=None) -> Tensor:
     """LayerNorm -> Linear; folded into ONE GEMM on GPU (see make_ln_fold).
 
     ``stats``: per-row (sum, sumsq) of x if already known (emitted by the
@@ -125,8 +128,9 @@ def ln_linear(x: Tensor, ln_weight: Tensor, ln_bias: Tensor, eps: float,
         w_p, s, c = fused
         if stats is None:
             stats = _hip.ln_stats(x.reshape(-1, x.shape[-1]))
-        return _hip.linear(x, w_p, bias, _ACT_CODE[act], residual, 0,
-                           stats, s, c, eps, stats_out)
+        out, _ = _hip.linear(x, w_p, bias, _ACT_CODE[act], residual, 0,
+                             stats, s, c, eps, False)
+        return out
     y = torch_ref.layernorm(x, ln_weight, ln_bias, eps)
     return torch_ref.linear(y, weight_nk, bias, act, residual)
 

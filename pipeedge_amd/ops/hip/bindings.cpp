@@ -4,15 +4,14 @@
 // every GPU tensor; there is no eager fallback on GPU (fail-loud contract).
 #include <torch/extension.h>
 
-torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
-                        c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual2d,
-                        int64_t tile = 0,
-                        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
-                        c10::optional<torch::Tensor> ln_s = c10::nullopt,
-                        c10::optional<torch::Tensor> ln_c = c10::nullopt,
-                        double ln_eps = 1e-12,
-                        c10::optional<torch::Tensor> stats_out = c10::nullopt);
+std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
+        torch::Tensor a2d, torch::Tensor b,
+        c10::optional<torch::Tensor> bias, int64_t act,
+        c10::optional<torch::Tensor> residual2d, int64_t tile = 0,
+        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
+        c10::optional<torch::Tensor> ln_s = c10::nullopt,
+        c10::optional<torch::Tensor> ln_c = c10::nullopt,
+        double ln_eps = 1e-12, bool want_stats = false);
 torch::Tensor ln_stats_bf16(torch::Tensor x);
 torch::Tensor debug_mfma_tile(torch::Tensor a, torch::Tensor b);
 torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor weight,
@@ -27,25 +26,24 @@ torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
 namespace {
 
 // reshape [.., K] -> [M, K] for the GEMM, restoring leading dims after
-torch::Tensor linear_op(torch::Tensor x, torch::Tensor w,
-                        c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual,
-                        int64_t tile = 0,
-                        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
-                        c10::optional<torch::Tensor> ln_s = c10::nullopt,
-                        c10::optional<torch::Tensor> ln_c = c10::nullopt,
-                        double ln_eps = 1e-12,
-                        c10::optional<torch::Tensor> stats_out = c10::nullopt) {
+std::tuple<torch::Tensor, torch::Tensor> linear_op(
+        torch::Tensor x, torch::Tensor w,
+        c10::optional<torch::Tensor> bias, int64_t act,
+        c10::optional<torch::Tensor> residual, int64_t tile = 0,
+        c10::optional<torch::Tensor> ln_stats = c10::nullopt,
+        c10::optional<torch::Tensor> ln_s = c10::nullopt,
+        c10::optional<torch::Tensor> ln_c = c10::nullopt,
+        double ln_eps = 1e-12, bool want_stats = false) {
     auto sizes = x.sizes().vec();
     const long K = sizes.back();
     auto x2d = x.reshape({-1, K});
     c10::optional<torch::Tensor> res2d;
     if (residual.has_value())
         res2d = residual->reshape({-1, w.size(0)});  // w is [N, K]
-    auto out = gemm_bf16(x2d, w, bias, act, res2d, tile, ln_stats, ln_s,
-                         ln_c, ln_eps, stats_out);
+    auto [out, stats] = gemm_bf16(x2d, w, bias, act, res2d, tile, ln_stats,
+                                  ln_s, ln_c, ln_eps, want_stats);
     sizes.back() = w.size(0);
-    return out.reshape(sizes);
+    return {out.reshape(sizes), stats};
 }
 
 torch::Tensor qkv_attention_op(torch::Tensor qkv, int64_t num_heads,
@@ -64,7 +62,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("act") = 0, py::arg("residual") = py::none(),
           py::arg("tile") = 0, py::arg("ln_stats") = py::none(),
           py::arg("ln_s") = py::none(), py::arg("ln_c") = py::none(),
-          py::arg("ln_eps") = 1e-12, py::arg("stats_out") = py::none());
+          py::arg("ln_eps") = 1e-12, py::arg("want_stats") = false);
     m.def("ln_stats", &ln_stats_bf16, py::arg("x"));
     m.def("layernorm", &layernorm_bf16, py::arg("x"), py::arg("weight"),
           py::arg("bias"), py::arg("eps") = 1e-6);

@@ -58,30 +58,47 @@ struct EpilogueArgs {
     // folded-LN input adjustment (see ops/__init__.py ln_linear): the GEMM
     // ran on RAW x against W' = W*ln_w; the epilogue applies
     //   out = r*acc - r*mu*ln_s[col] + ln_c[col]
-    // with per-row mu/r derived from ln_stats[row] = (sum, sumsq) over K.
-    const float* ln_stats;   // [M][2] or nullptr
+    // with per-row mu/r from the SUMMED partial stats
+    // ln_stats[p][row][2] (p < ln_parts), each partial written without
+    // atomics by one (column-block, column-wave) of the producing GEMM.
+    const float* ln_stats;   // [ln_parts][M][2] or nullptr
+    int ln_parts;
     const float* ln_s;       // [N] row-sums of W'
     const float* ln_c;       // [N] ln_b @ W^T
     float ln_eps;
     int ln_k;                // LN normalization length (= GEMM K)
-    // output-stats emission: per-row (sum, sumsq) of C accumulated with
-    // atomics into stats_out[M][2] (zeroed by the host before launch)
-    float* stats_out;
+    // output-stats emission: this block's (col-block by, col-wave) slice of
+    // stats_out[parts][M][2]; plain stores, no zeroing needed
+    float* stats_out;        // or nullptr
 };
 
 template <int MI, int NI>
 __device__ __forceinline__ void gemm_epilogue(
         const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
-        int wm, int wn, int lane) {
-    float row_sum[MI][4];
-    float row_sumsq[MI][4];
-    if (ep.stats_out != nullptr) {
+        int wm, int wn, int lane, int part_idx) {
+    // hoisted per-row LN coefficients (summed over partials)
+    float mu[MI][4], rstd[MI][4];
+    if (ep.ln_stats != nullptr) {
         #pragma unroll
-        for (int mi = 0; mi < MI; ++mi)
+        for (int mi = 0; mi < MI; ++mi) {
             #pragma unroll
-            for (int r = 0; r < 4; ++r)
-                row_sum[mi][r] = row_sumsq[mi][r] = 0.0f;
+            for (int r = 0; r < 4; ++r) {
+                const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
+                float s1 = 0.0f, s2 = 0.0f;
+                if (row < ep.M) {
+                    for (int p = 0; p < ep.ln_parts; ++p) {
+                        s1 += ep.ln_stats[((long)p * ep.M + row) * 2];
+                        s2 += ep.ln_stats[((long)p * ep.M + row) * 2 + 1];
+                    }
+                }
+                const float m = s1 / ep.ln_k;
+                mu[mi][r] = m;
+                rstd[mi][r] = rsqrtf(s2 / ep.ln_k - m * m + ep.ln_eps);
+            }
+        }
     }
+    float row_sum[MI][4] = {};
+    float row_sumsq[MI][4] = {};
     #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
         #pragma unroll
@@ -96,14 +113,8 @@ __device__ __forceinline__ void gemm_epilogue(
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= ep.M) continue;
                 float v = acc[mi][ni][r];
-                if (ep.ln_stats != nullptr) {
-                    const float s1 = ep.ln_stats[row * 2];
-                    const float s2 = ep.ln_stats[row * 2 + 1];
-                    const float mu = s1 / ep.ln_k;
-                    const float rstd = rsqrtf(s2 / ep.ln_k - mu * mu +
-                                              ep.ln_eps);
-                    v = rstd * v - rstd * mu * sv + cv;
-                }
+                if (ep.ln_stats != nullptr)
+                    v = rstd[mi][r] * v - rstd[mi][r] * mu[mi][r] * sv + cv;
                 v += bv;
                 if (ep.act == ACT_GELU) v = gelu_f(v);
                 else if (ep.act == ACT_TANH) v = tanhf(v);
@@ -118,18 +129,19 @@ __device__ __forceinline__ void gemm_epilogue(
         }
     }
     if (ep.stats_out != nullptr) {
-        // reduce across the 16 lanes sharing each C-fragment row, then one
-        // atomic pair per (wave, mi, r)
+        // reduce across the 16 lanes sharing each C-fragment row; ONE plain
+        // store per (row, this block's column-wave slice)
+        float* slice = ep.stats_out + (long)part_idx * ep.M * 2;
         #pragma unroll
         for (int mi = 0; mi < MI; ++mi) {
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float s1 = group16_reduce(row_sum[mi][r], SumOp());
-                float s2 = group16_reduce(row_sumsq[mi][r], SumOp());
+                const float s1 = group16_reduce(row_sum[mi][r], SumOp());
+                const float s2 = group16_reduce(row_sumsq[mi][r], SumOp());
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 if ((lane & 15) == 0 && row < ep.M) {
-                    atomicAdd(&ep.stats_out[row * 2], s1);
-                    atomicAdd(&ep.stats_out[row * 2 + 1], s2);
+                    slice[row * 2] = s1;
+                    slice[row * 2 + 1] = s2;
                 }
             }
         }
@@ -224,7 +236,8 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         __syncthreads();
     }
 
-    gemm_epilogue<MI, NI>(acc, ep, bm, bn, wm, wn, lane);
+    gemm_epilogue<MI, NI>(acc, ep, bm, bn, wm, wn, lane,
+                          (bn / TBN) * 2 + (wid & 1));
 }
 
 // ---------------------------------------------------------------------------
@@ -294,19 +307,19 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         __syncthreads();
     }
 
-    gemm_epilogue<2, 2>(acc, ep, bm, bn, wm, wn, lane);
+    gemm_epilogue<2, 2>(acc, ep, bm, bn, wm, wn, lane,
+                        (bn / BN) * 2 + (wid & 1));
 }
 
 }  // namespace
 
-torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
-                        c10::optional<torch::Tensor> bias, int64_t act,
-                        c10::optional<torch::Tensor> residual2d,
-                        int64_t tile,
-                        c10::optional<torch::Tensor> ln_stats,
-                        c10::optional<torch::Tensor> ln_s,
-                        c10::optional<torch::Tensor> ln_c, double ln_eps,
-                        c10::optional<torch::Tensor> stats_out) {
+std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
+        torch::Tensor a2d, torch::Tensor b,
+        c10::optional<torch::Tensor> bias, int64_t act,
+        c10::optional<torch::Tensor> residual2d, int64_t tile,
+        c10::optional<torch::Tensor> ln_stats,
+        c10::optional<torch::Tensor> ln_s,
+        c10::optional<torch::Tensor> ln_c, double ln_eps, bool want_stats) {
     TORCH_CHECK(a2d.is_cuda() && a2d.dtype() == torch::kBFloat16,
                 "gemm: A must be bf16 on GPU");
     TORCH_CHECK(b.dtype() == torch::kBFloat16, "gemm: B must be bf16");
@@ -336,26 +349,24 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
     auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
 
     EpilogueArgs ep{bias_p, res_p, cp, M, N, (int)act,
-                    nullptr, nullptr, nullptr, (float)ln_eps, K, nullptr};
+                    nullptr, 0, nullptr, nullptr, (float)ln_eps, K, nullptr};
     if (ln_stats.has_value()) {
         TORCH_CHECK(ln_s.has_value() && ln_c.has_value(),
                     "gemm: ln_stats requires ln_s and ln_c");
         TORCH_CHECK(ln_stats->dtype() == torch::kFloat32 &&
-                    ln_stats->numel() == 2 * M);
+                    ln_stats->dim() == 3 && ln_stats->size(1) == M &&
+                    ln_stats->size(2) == 2, "ln_stats must be [P, M, 2]");
         TORCH_CHECK(ln_s->numel() == N && ln_c->numel() == N);
         ep.ln_stats = ln_stats->data_ptr<float>();
+        ep.ln_parts = ln_stats->size(0);
         ep.ln_s = ln_s->data_ptr<float>();
         ep.ln_c = ln_c->data_ptr<float>();
     }
-    if (stats_out.has_value()) {
-        TORCH_CHECK(stats_out->dtype() == torch::kFloat32 &&
-                    stats_out->numel() == 2 * M);
-        ep.stats_out = stats_out->data_ptr<float>();
-    }
+    int t = 0;
     if (K % BK == 0) {
         // tile choice: empirically swept (tests/gpu sweep); override via
         // the `tile` arg (0 = auto heuristic)
-        int t = (int)tile;
+        t = (int)tile;
         if (t == 0) {
             // from the on-HW sweeps (tools/gemm_sweep.py,
             // tools/hotpath_sweep.py): XCD-swizzled 64x64 default; BK=128
@@ -366,6 +377,18 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
         }
         if (t == 4 && K % (2 * BK) != 0) t = 1;
         if (t == 6 && K % (2 * BK) != 0) t = 5;
+    }
+    torch::Tensor stats;
+    if (want_stats) {
+        const int bn_tile = t == 3 ? 128 : 64;
+        const int parts = ((N + bn_tile - 1) / bn_tile) * 2;
+        stats = torch::empty({parts, M, 2},
+                             a2d.options().dtype(torch::kFloat32));
+        ep.stats_out = stats.data_ptr<float>();
+    } else {
+        stats = torch::empty({0}, a2d.options().dtype(torch::kFloat32));
+    }
+    if (K % BK == 0) {
         if (t == 4 || t == 5 || t == 6) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             if (t == 4)
@@ -403,7 +426,7 @@ torch::Tensor gemm_bf16(torch::Tensor a2d, torch::Tensor b,
                            M, N, K, ep);
     }
     HIP_CHECK_LAST();
-    return out;
+    return {out, stats};
 }
 
 // debug helper: one 16x16x32 MFMA tile with the documented fragment layout -
