@@ -80,7 +80,7 @@ def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
            act: Optional[str] = None, residual: Optional[Tensor] = None,
            want_stats: bool = False):
     """GEMM + fused epilogue. With ``want_stats`` (GPU path) also returns
-    per-row partial (sum, sumsq) of the OUTPUT, f32 [parts, rows, 2] -
+    per-row partial (sum, sumsq) of the OUTPUT, f32 [rows, parts, 2] -
     written without atomics by each column block and summed by the next
     folded-LN GEMM (see ln_linear)."""
     if _use_hip(x):
@@ -93,11 +93,11 @@ def linear(x: Tensor, weight_nk: Tensor, bias: Optional[Tensor] = None,
 
 
 def ln_stats(x: Tensor) -> Tensor:
-    """Per-row (sum, sumsq) over the last dim, f32 [1, rows, 2]."""
+    """Per-row (sum, sumsq) over the last dim, f32 [rows, 1, 2]."""
     if _use_hip(x):
         return _hip.ln_stats(x.reshape(-1, x.shape[-1]))
     xf = x.reshape(-1, x.shape[-1]).float()
-    return torch.stack((xf.sum(-1), (xf * xf).sum(-1)), dim=-1).unsqueeze(0)
+    return torch.stack((xf.sum(-1), (xf * xf).sum(-1)), dim=-1).unsqueeze(1)
 
 
 def make_ln_fold(weight_nk: Tensor, ln_weight: Tensor, ln_bias: Tensor):

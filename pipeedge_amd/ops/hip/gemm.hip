@@ -61,7 +61,7 @@ struct EpilogueArgs {
     // with per-row mu/r from the SUMMED partial stats
     // ln_stats[p][row][2] (p < ln_parts), each partial written without
     // atomics by one (column-block, column-wave) of the producing GEMM.
-    const float* ln_stats;   // [ln_parts][M][2] or nullptr
+    const float* ln_stats;   // [M][ln_parts][2] or nullptr
     int ln_parts;
     const float* ln_s;       // [N] row-sums of W'
     const float* ln_c;       // [N] ln_b @ W^T
@@ -70,13 +70,17 @@ struct EpilogueArgs {
     // output-stats emission: this block's (col-block by, col-wave) slice of
     // stats_out[parts][M][2]; plain stores, no zeroing needed
     float* stats_out;        // or nullptr
+    int stats_parts;
 };
 
 template <int MI, int NI>
 __device__ __forceinline__ void gemm_epilogue(
         const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
         int wm, int wn, int lane, int part_idx) {
-    // hoisted per-row LN coefficients (summed over partials)
+    // hoisted per-row LN coefficients. Partials are [M][P][2]; the 16
+    // lanes sharing a C-fragment row split the P-sum (lane i takes
+    // p = i, i+16, ...), then a group16 shfl-reduce gives every lane the
+    // total - 2 loads + 8 shfl instead of 2*P scalar loads per lane.
     float mu[MI][4], rstd[MI][4];
     if (ep.ln_stats != nullptr) {
         #pragma unroll
@@ -86,11 +90,15 @@ __device__ __forceinline__ void gemm_epilogue(
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 float s1 = 0.0f, s2 = 0.0f;
                 if (row < ep.M) {
-                    for (int p = 0; p < ep.ln_parts; ++p) {
-                        s1 += ep.ln_stats[((long)p * ep.M + row) * 2];
-                        s2 += ep.ln_stats[((long)p * ep.M + row) * 2 + 1];
+                    const float* rp = ep.ln_stats +
+                        (long)row * ep.ln_parts * 2;
+                    for (int p = lane & 15; p < ep.ln_parts; p += 16) {
+                        s1 += rp[p * 2];
+                        s2 += rp[p * 2 + 1];
                     }
                 }
+                s1 = group16_reduce(s1, SumOp());
+                s2 = group16_reduce(s2, SumOp());
                 const float m = s1 / ep.ln_k;
                 mu[mi][r] = m;
                 rstd[mi][r] = rsqrtf(s2 / ep.ln_k - m * m + ep.ln_eps);
@@ -130,8 +138,7 @@ __device__ __forceinline__ void gemm_epilogue(
     }
     if (ep.stats_out != nullptr) {
         // reduce across the 16 lanes sharing each C-fragment row; ONE plain
-        // store per (row, this block's column-wave slice)
-        float* slice = ep.stats_out + (long)part_idx * ep.M * 2;
+        // store per (row, this block's column-wave slice); layout [M][P][2]
         #pragma unroll
         for (int mi = 0; mi < MI; ++mi) {
             #pragma unroll
@@ -140,8 +147,10 @@ __device__ __forceinline__ void gemm_epilogue(
                 const float s2 = group16_reduce(row_sumsq[mi][r], SumOp());
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 if ((lane & 15) == 0 && row < ep.M) {
-                    slice[row * 2] = s1;
-                    slice[row * 2 + 1] = s2;
+                    float* slot = ep.stats_out +
+                        ((long)row * ep.stats_parts + part_idx) * 2;
+                    slot[0] = s1;
+                    slot[1] = s2;
                 }
             }
         }
@@ -349,16 +358,17 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
     auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
 
     EpilogueArgs ep{bias_p, res_p, cp, M, N, (int)act,
-                    nullptr, 0, nullptr, nullptr, (float)ln_eps, K, nullptr};
+                    nullptr, 0, nullptr, nullptr, (float)ln_eps, K, nullptr,
+                    0};
     if (ln_stats.has_value()) {
         TORCH_CHECK(ln_s.has_value() && ln_c.has_value(),
                     "gemm: ln_stats requires ln_s and ln_c");
         TORCH_CHECK(ln_stats->dtype() == torch::kFloat32 &&
-                    ln_stats->dim() == 3 && ln_stats->size(1) == M &&
-                    ln_stats->size(2) == 2, "ln_stats must be [P, M, 2]");
+                    ln_stats->dim() == 3 && ln_stats->size(0) == M &&
+                    ln_stats->size(2) == 2, "ln_stats must be [M, P, 2]");
         TORCH_CHECK(ln_s->numel() == N && ln_c->numel() == N);
         ep.ln_stats = ln_stats->data_ptr<float>();
-        ep.ln_parts = ln_stats->size(0);
+        ep.ln_parts = ln_stats->size(1);
         ep.ln_s = ln_s->data_ptr<float>();
         ep.ln_c = ln_c->data_ptr<float>();
     }
@@ -382,9 +392,10 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
     if (want_stats) {
         const int bn_tile = t == 3 ? 128 : 64;
         const int parts = ((N + bn_tile - 1) / bn_tile) * 2;
-        stats = torch::empty({parts, M, 2},
+        stats = torch::empty({M, parts, 2},
                              a2d.options().dtype(torch::kFloat32));
         ep.stats_out = stats.data_ptr<float>();
+        ep.stats_parts = parts;
     } else {
         stats = torch::empty({0}, a2d.options().dtype(torch::kFloat32));
     }

@@ -305,7 +305,7 @@ torch::Tensor ln_stats_bf16(torch::Tensor x) {
     auto x_c = x.contiguous();
     const int D = x.size(-1);
     const long rows = x.numel() / D;
-    auto stats = torch::empty({1, rows, 2},
+    auto stats = torch::empty({rows, 1, 2},
                               x.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(ln_stats_kernel, dim3((rows + 3) / 4), dim3(256), 0,
                        pe_current_stream(),

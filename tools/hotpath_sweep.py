@@ -55,3 +55,17 @@ for rows, d in [(1576,768),(1576,1024),(4096,768)]:
     assert (got.float()-ref).abs().max().item() < 0.05, (rows,d)
     us = bench(lambda: hip.layernorm(x,w,bb,1e-6))
     print(f"LN {rows}x{d}: {us:.2f} us  ({rows*d*2*2/1e3/us:.0f} GB/s)")
+
+
+print("== folded-LN A/B")
+m, n, k = 1576, 768, 768
+a = (torch.randn(m,k)*0.3).bfloat16().cuda()
+b = (torch.randn(n,k)*0.1).bfloat16().cuda()
+print(f"plain linear:      {bench(lambda: hip.linear(a,b)):7.1f} us")
+print(f"linear+want_stats: {bench(lambda: hip.linear(a,b,want_stats=True)):7.1f} us")
+_, stats = hip.linear(a, b, want_stats=True)
+bq = (torch.randn(2304,768)*0.1).bfloat16().cuda()
+s = bq.float().sum(1).contiguous(); c = torch.zeros(2304, device="cuda")
+print(f"qkv plain:         {bench(lambda: hip.linear(a,bq)):7.1f} us")
+print(f"qkv folded-ln:     {bench(lambda: hip.linear(a,bq,ln_stats=stats,ln_s=s,ln_c=c)):7.1f} us")
+print(f"ln_stats kernel:   {bench(lambda: hip.ln_stats(a)):7.1f} us")
