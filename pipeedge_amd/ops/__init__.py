@@ -136,8 +136,14 @@ def ln_linear(x: Tensor, ln_weight: Tensor, ln_bias: Tensor, eps: float,
 
 
 def fused_ln_active(x: Tensor) -> bool:
-    """Whether the folded-LN GEMM path will be used for this tensor."""
-    return x.is_cuda and not force_torch()
+    """Whether the folded-LN GEMM path is used for this tensor.
+
+    Default OFF: end-to-end A/B on MI355X measured the folded path at net
+    ~parity-to-slower vs the standalone LN kernels (the LN savings are offset
+    by partial-stats traffic in the consumer epilogues); enable with
+    PIPEEDGE_FOLDED_LN=1 to experiment."""
+    return (x.is_cuda and not force_torch()
+            and os.environ.get("PIPEEDGE_FOLDED_LN", "0") == "1")
 
 
 def attention_core(q: Tensor, k: Tensor, v: Tensor, scale: Optional[float] = None,

@@ -67,41 +67,54 @@ struct EpilogueArgs {
     const float* ln_c;       // [N] ln_b @ W^T
     float ln_eps;
     int ln_k;                // LN normalization length (= GEMM K)
+    float ln_rk;             // 1.0f / ln_k (full f32 div is ~10 instrs)
     // output-stats emission: this block's (col-block by, col-wave) slice of
     // stats_out[parts][M][2]; plain stores, no zeroing needed
     float* stats_out;        // or nullptr
     int stats_parts;
 };
 
+// cooperative per-block LN-coefficient pass: thread t < TBM sums row
+// (bm + t)'s partials and writes (mu, rstd) into the (reused) LDS tile -
+// one thread per row, loads pipeline in parallel, no cross-lane shuffles,
+// no divisions in the per-element epilogue loops.
+template <int TBM>
+__device__ __forceinline__ void ln_coeffs_to_lds(
+        const EpilogueArgs& ep, int bm, int tid, float* lnbuf) {
+    __syncthreads();  // k-loop reads of the LDS tile are done
+    if (tid < TBM) {
+        const int row = bm + tid;
+        float mu = 0.0f, rstd = 1.0f;
+        if (row < ep.M) {
+            const float* rp = ep.ln_stats + (long)row * ep.ln_parts * 2;
+            float s1 = 0.0f, s2 = 0.0f;
+            for (int p = 0; p < ep.ln_parts; ++p) {
+                s1 += rp[p * 2];
+                s2 += rp[p * 2 + 1];
+            }
+            mu = s1 * ep.ln_rk;
+            rstd = rsqrtf(s2 * ep.ln_rk - mu * mu + ep.ln_eps);
+        }
+        lnbuf[tid * 2] = mu;
+        lnbuf[tid * 2 + 1] = rstd;
+    }
+    __syncthreads();
+}
+
 template <int MI, int NI>
 __device__ __forceinline__ void gemm_epilogue(
         const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
-        int wm, int wn, int lane, int part_idx) {
-    // hoisted per-row LN coefficients. Partials are [M][P][2]; the 16
-    // lanes sharing a C-fragment row split the P-sum (lane i takes
-    // p = i, i+16, ...), then a group16 shfl-reduce gives every lane the
-    // total - 2 loads + 8 shfl instead of 2*P scalar loads per lane.
+        int wm, int wn, int lane, int part_idx, const float* lnbuf) {
+    // per-row LN coefficients from the block's LDS buffer (broadcast reads)
     float mu[MI][4], rstd[MI][4];
     if (ep.ln_stats != nullptr) {
         #pragma unroll
         for (int mi = 0; mi < MI; ++mi) {
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
-                float s1 = 0.0f, s2 = 0.0f;
-                if (row < ep.M) {
-                    const float* rp = ep.ln_stats +
-                        (long)row * ep.ln_parts * 2;
-                    for (int p = lane & 15; p < ep.ln_parts; p += 16) {
-                        s1 += rp[p * 2];
-                        s2 += rp[p * 2 + 1];
-                    }
-                }
-                s1 = group16_reduce(s1, SumOp());
-                s2 = group16_reduce(s2, SumOp());
-                const float m = s1 / ep.ln_k;
-                mu[mi][r] = m;
-                rstd[mi][r] = rsqrtf(s2 / ep.ln_k - m * m + ep.ln_eps);
+                const int lrow = wm + mi * 16 + (lane >> 4) * 4 + r;
+                mu[mi][r] = lnbuf[lrow * 2];
+                rstd[mi][r] = lnbuf[lrow * 2 + 1];
             }
         }
     }
@@ -245,8 +258,11 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         __syncthreads();
     }
 
+    float* lnbuf = reinterpret_cast<float*>(&smem[0][0][0]);
+    if (ep.ln_stats != nullptr)
+        ln_coeffs_to_lds<TBM>(ep, bm, tid, lnbuf);
     gemm_epilogue<MI, NI>(acc, ep, bm, bn, wm, wn, lane,
-                          (bn / TBN) * 2 + (wid & 1));
+                          (bn / TBN) * 2 + (wid & 1), lnbuf);
 }
 
 // ---------------------------------------------------------------------------
@@ -316,8 +332,11 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         __syncthreads();
     }
 
+    float* lnbuf = reinterpret_cast<float*>(&As[0][0]);
+    if (ep.ln_stats != nullptr)
+        ln_coeffs_to_lds<BM>(ep, bm, tid, lnbuf);
     gemm_epilogue<2, 2>(acc, ep, bm, bn, wm, wn, lane,
-                        (bn / BN) * 2 + (wid & 1));
+                        (bn / BN) * 2 + (wid & 1), lnbuf);
 }
 
 }  // namespace
@@ -358,8 +377,8 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
     auto* cp = reinterpret_cast<__bf16*>(out.data_ptr());
 
     EpilogueArgs ep{bias_p, res_p, cp, M, N, (int)act,
-                    nullptr, 0, nullptr, nullptr, (float)ln_eps, K, nullptr,
-                    0};
+                    nullptr, 0, nullptr, nullptr, (float)ln_eps, K,
+                    1.0f / K, nullptr, 0};
     if (ln_stats.has_value()) {
         TORCH_CHECK(ln_s.has_value() && ln_c.has_value(),
                     "gemm: ln_stats requires ln_s and ln_c");

@@ -21,7 +21,7 @@ for name,m,n,k in shapes:
     ref = (a.float() @ b.float().t())
     row = f"{name:6s}"
     for t in (1,2,4,5,6):
-        out = hip.linear(a,b,tile=t)
+        out = hip.linear(a,b,tile=t)[0]
         err = (out.float()-ref).abs().max().item()
         assert err < 1.0, (name, t, err)
         us = bench(lambda t=t: hip.linear(a,b,tile=t))
