@@ -101,13 +101,13 @@ __device__ __forceinline__ void ln_coeffs_to_lds(
     __syncthreads();
 }
 
-template <int MI, int NI>
+template <int MI, int NI, int EPM>  // EPM: 0 plain, 1 ln-in, 2 stats-out
 __device__ __forceinline__ void gemm_epilogue(
         const f32x4 (&acc)[MI][NI], const EpilogueArgs& ep, int bm, int bn,
         int wm, int wn, int lane, int part_idx, const float* lnbuf) {
     // per-row LN coefficients from the block's LDS buffer (broadcast reads)
     float mu[MI][4], rstd[MI][4];
-    if (ep.ln_stats != nullptr) {
+    if constexpr (EPM == 1) {
         #pragma unroll
         for (int mi = 0; mi < MI; ++mi) {
             #pragma unroll
@@ -118,8 +118,8 @@ __device__ __forceinline__ void gemm_epilogue(
             }
         }
     }
-    float row_sum[MI][4] = {};
-    float row_sumsq[MI][4] = {};
+    float row_sum[EPM == 2 ? MI : 1][4] = {};
+    float row_sumsq[EPM == 2 ? MI : 1][4] = {};
     #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
         #pragma unroll
@@ -127,14 +127,14 @@ __device__ __forceinline__ void gemm_epilogue(
             const int col = bn + wn + ni * 16 + (lane & 15);
             if (col >= ep.N) continue;
             const float bv = ep.bias != nullptr ? bf2f(ep.bias[col]) : 0.0f;
-            const float sv = ep.ln_stats != nullptr ? ep.ln_s[col] : 0.0f;
-            const float cv = ep.ln_stats != nullptr ? ep.ln_c[col] : 0.0f;
+            const float sv = EPM == 1 ? ep.ln_s[col] : 0.0f;
+            const float cv = EPM == 1 ? ep.ln_c[col] : 0.0f;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = bm + wm + mi * 16 + (lane >> 4) * 4 + r;
                 if (row >= ep.M) continue;
                 float v = acc[mi][ni][r];
-                if (ep.ln_stats != nullptr)
+                if constexpr (EPM == 1)
                     v = rstd[mi][r] * v - rstd[mi][r] * mu[mi][r] * sv + cv;
                 v += bv;
                 if (ep.act == ACT_GELU) v = gelu_f(v);
@@ -142,14 +142,14 @@ __device__ __forceinline__ void gemm_epilogue(
                 if (ep.residual != nullptr)
                     v += bf2f(ep.residual[(long)row * ep.N + col]);
                 ep.C[(long)row * ep.N + col] = f2bf(v);
-                if (ep.stats_out != nullptr) {
+                if constexpr (EPM == 2) {
                     row_sum[mi][r] += v;
                     row_sumsq[mi][r] += v * v;
                 }
             }
         }
     }
-    if (ep.stats_out != nullptr) {
+    if constexpr (EPM == 2) {
         // reduce across the 16 lanes sharing each C-fragment row; ONE plain
         // store per (row, this block's column-wave slice); layout [M][P][2]
         #pragma unroll
@@ -175,7 +175,7 @@ __device__ __forceinline__ void gemm_epilogue(
 // each wave an MIxNI grid of 16x16 fragments. <2,2> = 64x64 (small-N
 // shapes need the block count), <4,4> = 128x128 (fc1/qkv class).
 // ---------------------------------------------------------------------------
-template <int MI, int NI, int KT = 1, bool XCD = false>
+template <int MI, int NI, int KT = 1, bool XCD = false, int EPM = 0>
 __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
@@ -259,15 +259,16 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
     }
 
     float* lnbuf = reinterpret_cast<float*>(&smem[0][0][0]);
-    if (ep.ln_stats != nullptr)
+    if constexpr (EPM == 1)
         ln_coeffs_to_lds<TBM>(ep, bm, tid, lnbuf);
-    gemm_epilogue<MI, NI>(acc, ep, bm, bn, wm, wn, lane,
-                          (bn / TBN) * 2 + (wid & 1), lnbuf);
+    gemm_epilogue<MI, NI, EPM>(acc, ep, bm, bn, wm, wn, lane,
+                               (bn / TBN) * 2 + (wid & 1), lnbuf);
 }
 
 // ---------------------------------------------------------------------------
 // generic path: any K (padded LDS, plain vectorized staging)
 // ---------------------------------------------------------------------------
+template <int EPM = 0>
 __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
@@ -333,10 +334,10 @@ __global__ __launch_bounds__(256) void gemm_bf16_generic_kernel(
     }
 
     float* lnbuf = reinterpret_cast<float*>(&As[0][0]);
-    if (ep.ln_stats != nullptr)
+    if constexpr (EPM == 1)
         ln_coeffs_to_lds<BM>(ep, bm, tid, lnbuf);
-    gemm_epilogue<2, 2>(acc, ep, bm, bn, wm, wn, lane,
-                        (bn / BN) * 2 + (wid & 1), lnbuf);
+    gemm_epilogue<2, 2, EPM>(acc, ep, bm, bn, wm, wn, lane,
+                             (bn / BN) * 2 + (wid & 1), lnbuf);
 }
 
 }  // namespace
@@ -418,43 +419,55 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
     } else {
         stats = torch::empty({0}, a2d.options().dtype(torch::kFloat32));
     }
+    const int epm = ep.ln_stats != nullptr ? 1
+                    : (ep.stats_out != nullptr ? 2 : 0);
+#define PE_LAUNCH_GLDS(MI_, NI_, KT_, XCD_, GRID_)                           \
+    do {                                                                     \
+        if (epm == 1)                                                        \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds_kernel<MI_, NI_, KT_, XCD_, 1>), GRID_,      \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
+        else if (epm == 2)                                                   \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds_kernel<MI_, NI_, KT_, XCD_, 2>), GRID_,      \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
+        else                                                                 \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds_kernel<MI_, NI_, KT_, XCD_, 0>), GRID_,      \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
+    } while (0)
     if (K % BK == 0) {
         if (t == 4 || t == 5 || t == 6) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-            if (t == 4)
-                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2>), grid,
-                                   dim3(256), 0, pe_current_stream(), ap, bp,
-                                   cp, M, N, K, ep);
-            else if (t == 5)
-                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 1, true>),
-                                   grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, cp, M, N, K, ep);
-            else
-                hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2, 2, true>),
-                                   grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, cp, M, N, K, ep);
+            if (t == 4) PE_LAUNCH_GLDS(2, 2, 2, false, grid);
+            else if (t == 5) PE_LAUNCH_GLDS(2, 2, 1, true, grid);
+            else PE_LAUNCH_GLDS(2, 2, 2, true, grid);
         } else if (t == 3) {
             dim3 grid((M + 127) / 128, (N + 127) / 128);
-            hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 4>), grid,
-                               dim3(256), 0, pe_current_stream(), ap, bp,
-                               cp, M, N, K, ep);
+            PE_LAUNCH_GLDS(4, 4, 1, false, grid);
         } else if (t == 2) {
             dim3 grid((M + 127) / 128, (N + 63) / 64);
-            hipLaunchKernelGGL((gemm_bf16_glds_kernel<4, 2>), grid,
-                               dim3(256), 0, pe_current_stream(), ap, bp,
-                               cp, M, N, K, ep);
+            PE_LAUNCH_GLDS(4, 2, 1, false, grid);
         } else {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-            hipLaunchKernelGGL((gemm_bf16_glds_kernel<2, 2>), grid,
-                               dim3(256), 0, pe_current_stream(), ap, bp,
-                               cp, M, N, K, ep);
+            PE_LAUNCH_GLDS(2, 2, 1, false, grid);
         }
     } else {
         dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-        hipLaunchKernelGGL(gemm_bf16_generic_kernel, grid, dim3(256), 0,
-                           pe_current_stream(), ap, bp, cp,
-                           M, N, K, ep);
+        if (epm == 1)
+            hipLaunchKernelGGL((gemm_bf16_generic_kernel<1>), grid, dim3(256),
+                               0, pe_current_stream(), ap, bp, cp, M, N, K,
+                               ep);
+        else if (epm == 2)
+            hipLaunchKernelGGL((gemm_bf16_generic_kernel<2>), grid, dim3(256),
+                               0, pe_current_stream(), ap, bp, cp, M, N, K,
+                               ep);
+        else
+            hipLaunchKernelGGL((gemm_bf16_generic_kernel<0>), grid, dim3(256),
+                               0, pe_current_stream(), ap, bp, cp, M, N, K,
+                               ep);
     }
+#undef PE_LAUNCH_GLDS
     HIP_CHECK_LAST();
     return {out, stats};
 }
