@@ -67,38 +67,62 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
     float l_run[4] = {};
 
-    const int n_tiles = (S + KVT - 1) / KVT;
-    for (int t = 0; t < n_tiles; ++t) {
+    // T14 async-stage split (guide section 6 G15): tile t+1's K/V global
+    // loads issue BEFORE tile t's compute so HBM latency hides under the
+    // MFMA phase; the LDS writes happen after the next barrier.
+    constexpr int PIECES = KVT * DHP / 8 / 256;  // vec8 pieces per thread
+    bf16x8 k_reg[PIECES], v_reg[PIECES];
+
+    auto load_tile = [&](int t) {
         const int s0 = t * KVT;
-        __syncthreads();
-        // ---- stage K tile (natural) + V tile (transposed) ----
-        for (int piece = tid; piece < KVT * DHP / 8; piece += 256) {
+        #pragma unroll
+        for (int i = 0; i < PIECES; ++i) {
+            const int piece = tid + i * 256;
             const int r = piece / (DHP / 8);
             const int c = (piece % (DHP / 8)) * 8;
             const int gs = s0 + r;
-            bf16x8 kv_ = {};
-            bf16x8 vv = {};
+            k_reg[i] = bf16x8{};
+            v_reg[i] = bf16x8{};
             if (gs < S && c < dh) {
                 if (c + 7 < dh) {
-                    kv_ = *reinterpret_cast<const bf16x8*>(
+                    k_reg[i] = *reinterpret_cast<const bf16x8*>(
                         k_base + (long)gs * row_stride + c);
-                    vv = *reinterpret_cast<const bf16x8*>(
+                    v_reg[i] = *reinterpret_cast<const bf16x8*>(
                         v_base + (long)gs * row_stride + c);
                 } else {
                     for (int j = 0; j < 8 && c + j < dh; ++j) {
-                        kv_[j] = k_base[(long)gs * row_stride + c + j];
-                        vv[j] = v_base[(long)gs * row_stride + c + j];
+                        k_reg[i][j] = k_base[(long)gs * row_stride + c + j];
+                        v_reg[i][j] = v_base[(long)gs * row_stride + c + j];
                     }
                 }
             }
-            *reinterpret_cast<bf16x8*>(&Ks[r][c]) = kv_;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) Vs[c + j][r] = vv[j];
         }
-        __syncthreads();
+    };
+    auto write_tile = [&]() {
+        #pragma unroll
+        for (int i = 0; i < PIECES; ++i) {
+            const int piece = tid + i * 256;
+            const int r = piece / (DHP / 8);
+            const int c = (piece % (DHP / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(&Ks[r][c]) = k_reg[i];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) Vs[c + j][r] = v_reg[i][j];
+        }
+    };
+
+    const int n_tiles = (S + KVT - 1) / KVT;
+    load_tile(0);
+    for (int t = 0; t < n_tiles; ++t) {
+        const int s0 = t * KVT;
+        __syncthreads();          // previous tile's LDS reads complete
+        write_tile();
+        __syncthreads();          // staging visible
+        if (t + 1 < n_tiles)
+            load_tile(t + 1);     // flies under this tile's compute
 
         // ---- scores: S16x64 = Q16xDHP . K^T ----
         f32x4 sfrag[KVT / 16] = {};
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kk = 0; kk < DHP; kk += 32) {
             const int kbase = kk + (lane >> 4) * 8;
@@ -111,6 +135,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
                 sfrag[nf] = mfma_16x16x32_bf16(a, bb, sfrag[nf]);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
         // scale + mask invalid kv columns
         #pragma unroll
         for (int nf = 0; nf < KVT / 16; ++nf) {
@@ -154,6 +179,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
                     f2bf(p_local[nf][r]);
 
         // ---- PV accumulate: O16xDHP += P16x64 . V64xDHP ----
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kk = 0; kk < KVT; kk += 32) {
             const int kbase = kk + (lane >> 4) * 8;
@@ -166,6 +192,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
                 o_acc[nfo] = mfma_16x16x32_bf16(a, bb, o_acc[nfo]);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- epilogue: normalize and store ctx rows ----

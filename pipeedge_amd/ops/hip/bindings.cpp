@@ -22,6 +22,11 @@ torch::Tensor add_layernorm_bf16(torch::Tensor x, torch::Tensor residual,
 torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
                                  double scale);
 torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
+std::vector<torch::Tensor> quant_encode_bf16(torch::Tensor x, int64_t bit,
+                                             int64_t pad_words);
+torch::Tensor quant_decode_bf16(torch::Tensor comm, torch::Tensor scale,
+                                torch::Tensor shift, int64_t bit,
+                                std::vector<int64_t> image_shape);
 
 namespace {
 
@@ -74,5 +79,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("attn_mask") = py::none());
     m.def("patchify", &patchify_bf16, py::arg("pixels"),
           py::arg("patch_size"));
+    m.def("quant_encode", &quant_encode_bf16, py::arg("x"), py::arg("bit"),
+          py::arg("pad_words") = 0);
+    m.def("quant_decode", &quant_decode_bf16, py::arg("comm"),
+          py::arg("scale"), py::arg("shift"), py::arg("bit"),
+          py::arg("image_shape"));
     m.def("_debug_mfma_tile", &debug_mfma_tile);
 }

@@ -146,15 +146,51 @@ def tensor_decode(encodings: List[Tensor]) -> Tensor:
     return (x * scale.float() + shift.float()).view(shape_list)
 
 
+def _hip_ops():
+    """The gfx950 kernel extension, or None on CPU / when forced off."""
+    from .. import ops as _ops
+    if _ops.force_torch() or not _ops.hip_available():
+        return None
+    from ..ops import hip_loader
+    return hip_loader.load()
+
+
 def tensor_encode_outerdim(batched_tensor: Tensor, quant_bit: int,
                            pad_words: Optional[int] = None) -> List[Tensor]:
     """Encode each element over the microbatch dim and stack
-    (reference basic_op.py:166-170)."""
+    (reference basic_op.py:166-170). On GPU the whole microbatch is encoded
+    by the fused HIP min/max + bit-pack kernels (quantpack.hip)."""
+    if batched_tensor.is_cuda and quant_bit > 0:
+        hip = _hip_ops()
+        if hip is None:
+            raise RuntimeError("quant encode on GPU requires the HIP "
+                               "extension (or PIPEEDGE_AMD_FORCE_TORCH=1)")
+        u = batched_tensor.shape[0]
+        img_shape = list(batched_tensor.shape[1:])
+        comm, scale, shift = hip.quant_encode(
+            batched_tensor.to(torch.bfloat16).contiguous(), quant_bit,
+            pad_words or 0)
+        shape = torch.tensor(img_shape, dtype=torch.int32,
+                             device=batched_tensor.device).unsqueeze(0).expand(u, -1).contiguous()
+        bits = torch.full((u,), quant_bit, dtype=torch.int8,
+                          device=batched_tensor.device)
+        return [comm, shape, scale, shift, bits]
     lists = [tensor_encode(t, quant_bit, pad_words) for t in batched_tensor]
     return [torch.stack(group, 0) for group in zip(*lists)]
 
 
 def tensor_decode_outerdim(batched_encodings: List[Tensor]) -> Tensor:
     """Decode a stacked 5-tensor group (reference basic_op.py:173-176)."""
+    comm, shape, scale, shift, bits = batched_encodings
+    if comm.is_cuda:
+        quant_bit = int(bits.view(-1)[0].item())
+        if quant_bit > 0:
+            hip = _hip_ops()
+            if hip is None:
+                raise RuntimeError("quant decode on GPU requires the HIP "
+                                   "extension (or PIPEEDGE_AMD_FORCE_TORCH=1)")
+            img_shape = [int(s) for s in shape.view(shape.shape[0], -1)[0]]
+            return hip.quant_decode(comm.contiguous(), scale, shift,
+                                    quant_bit, img_shape)
     tensors = [tensor_decode(list(enc)) for enc in zip(*batched_encodings)]
     return torch.stack(tensors, 0)

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "layernorm.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
         os.path.join(HIP_DIR, "patchify.hip"),
+        os.path.join(HIP_DIR, "quantpack.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

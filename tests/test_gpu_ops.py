@@ -224,3 +224,31 @@ def test_deit_forward_hip_vs_cpu_fp32():
     cos = torch.nn.functional.cosine_similarity(
         out.float().cpu().flatten(), ref.flatten(), dim=0)
     assert cos.item() > 0.98, cos.item()
+
+
+@pytest.mark.parametrize("bit", [2, 4, 6, 8, 16])
+def test_quant_hip_kernels_match_torch_path(bit):
+    """Fused HIP quant encode/decode vs the torch implementation of the same
+    wire format (pipeedge_amd.quant CPU path)."""
+    import math
+    from pipeedge_amd import quant
+    torch.manual_seed(bit)
+    x = torch.randn(4, 197, 768)
+    pad_words = math.ceil(197 * 768 / (32 // bit))
+    # CPU/torch reference wire
+    ref_enc = quant.tensor_encode_outerdim(x, bit, pad_words=pad_words)
+    ref_dec = quant.tensor_decode_outerdim(ref_enc)
+    # HIP wire
+    enc = quant.tensor_encode_outerdim(x.bfloat16().cuda(), bit,
+                                       pad_words=pad_words)
+    assert enc[0].dtype == torch.uint8 and enc[0].is_cuda
+    assert tuple(enc[0].shape) == tuple(ref_enc[0].shape)
+    dec = quant.tensor_decode_outerdim(enc)
+    # both decodes approximate x within the quantization error bound
+    levels = (1 << bit) - 1
+    bound = 1.2 / levels + 0.02
+    assert (dec.float().cpu() - x).abs().max().item() <= bound
+    assert (dec.float().cpu() - ref_dec).abs().max().item() <= bound
+    # cross-decode: HIP wire decoded by the torch path matches
+    cpu_dec = quant.tensor_decode_outerdim([t.cpu() for t in enc])
+    assert (cpu_dec - dec.float().cpu()).abs().max().item() < 0.02
