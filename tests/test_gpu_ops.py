@@ -244,11 +244,14 @@ def test_quant_hip_kernels_match_torch_path(bit):
     assert enc[0].dtype == torch.uint8 and enc[0].is_cuda
     assert tuple(enc[0].shape) == tuple(ref_enc[0].shape)
     dec = quant.tensor_decode_outerdim(enc)
-    # both decodes approximate x within the quantization error bound
+    # bounds: +-0.5 level on the per-image scale, plus bf16 input/output
+    # rounding (the HIP path quantizes the bf16-cast tensor; the reference
+    # path quantizes fp32 - grids differ by up to 1 level)
     levels = (1 << bit) - 1
-    bound = 1.2 / levels + 0.02
+    scale = enc[2].max().item()
+    bound = 0.6 * scale / levels + 0.05
     assert (dec.float().cpu() - x).abs().max().item() <= bound
-    assert (dec.float().cpu() - ref_dec).abs().max().item() <= bound
-    # cross-decode: HIP wire decoded by the torch path matches
+    assert (dec.float().cpu() - ref_dec).abs().max().item() <= 2 * bound
+    # cross-decode: HIP wire decoded by the torch path matches HIP decode
     cpu_dec = quant.tensor_decode_outerdim([t.cpu() for t in enc])
-    assert (cpu_dec - dec.float().cpu()).abs().max().item() < 0.02
+    assert (cpu_dec - dec.float().cpu()).abs().max().item() < 0.05
