@@ -185,7 +185,8 @@ def test_quant_roundtrip_on_gpu():
     enc = quant.tensor_encode_outerdim(x, 8)
     assert enc[0].is_cuda
     dec = quant.tensor_decode_outerdim(enc)
-    assert (dec - x).abs().max().item() <= 0.5 / 255 + 1e-6
+    # HIP path quantizes the bf16-cast input and emits bf16: ~2 extra ulps
+    assert (dec.float() - x).abs().max().item() <= 0.5 / 255 + 0.008
 
 
 def test_bert_forward_hip_vs_cpu_fp32():
