@@ -46,14 +46,23 @@ __device__ __forceinline__ float gelu_f(float x) {
 #define ACT_GELU 1
 #define ACT_TANH 2
 
-// reduce over the 16-lane groups that share a C-fragment row set
-// (lanes l and l^m for m in {1,2,4,8} have the same l>>4)
+// reduce over the 16-lane groups that share a C-fragment row set.
+// Implemented with DPP row rotations (one VALU op per step: a DPP "row" is
+// exactly 16 lanes) - __shfl_xor compiles to ds_bpermute (LDS-path, ~10x
+// the cost) on gfx950. Rotations suffice for commutative reductions.
 template <typename Op>
 __device__ __forceinline__ float group16_reduce(float v, Op op) {
-    v = op(v, __shfl_xor(v, 1));
-    v = op(v, __shfl_xor(v, 2));
-    v = op(v, __shfl_xor(v, 4));
-    v = op(v, __shfl_xor(v, 8));
+    #define PE_DPP_STEP(ctrl)                                               \
+        {                                                                   \
+            const int t_ = __builtin_amdgcn_update_dpp(                     \
+                0, __builtin_bit_cast(int, v), (ctrl), 0xf, 0xf, true);     \
+            v = op(v, __builtin_bit_cast(float, t_));                       \
+        }
+    PE_DPP_STEP(0x128)  // row_ror:8
+    PE_DPP_STEP(0x124)  // row_ror:4
+    PE_DPP_STEP(0x122)  // row_ror:2
+    PE_DPP_STEP(0x121)  // row_ror:1
+    #undef PE_DPP_STEP
     return v;
 }
 
