@@ -44,6 +44,16 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
         return reinterpret_cast<__bf16*>(
             reinterpret_cast<char*>(&Vs[0][0]) + byte);
     };
+    // Ps: same story at smaller scale - the C-layout P writes put lane
+    // groups 0/2 and 1/3 on shared banks (2-way); XOR bank-bit 3 by the
+    // row's upper half separates them and permutes-but-keeps-distinct the
+    // 16-lane b128 A-fragment reads.
+    auto ps_at = [&](int w, int row, int col) -> __bf16* {
+        int byte = ((w * 16 + row) * (KVT + PAD) + col) * 2;
+        byte ^= ((row >> 3) & 1) << 5;
+        return reinterpret_cast<__bf16*>(
+            reinterpret_cast<char*>(&Ps[0][0][0]) + byte);
+    };
 
     const int bh = blockIdx.x;
     const int b = bh / H;
@@ -188,7 +198,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
         for (int nf = 0; nf < KVT / 16; ++nf)
             #pragma unroll
             for (int r = 0; r < 4; ++r)
-                Ps[wid][(lane >> 4) * 4 + r][nf * 16 + (lane & 15)] =
+                *ps_at(wid, (lane >> 4) * 4 + r, nf * 16 + (lane & 15)) =
                     f2bf(p_local[nf][r]);
 
         // ---- PV accumulate: O16xDHP += P16x64 . V64xDHP ----
@@ -197,7 +207,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
         for (int kk = 0; kk < KVT; kk += 32) {
             const int kbase = kk + (lane >> 4) * 8;
             const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                &Ps[wid][lane & 15][kbase]);
+                ps_at(wid, lane & 15, kbase));
             #pragma unroll
             for (int nfo = 0; nfo < DHP / 16; ++nfo) {
                 const bf16x8 bb = *reinterpret_cast<const bf16x8*>(
