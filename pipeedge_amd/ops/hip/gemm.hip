@@ -32,12 +32,18 @@ __device__ __forceinline__ void glds_b128(const void* gsrc, void* lds_dst) {
         (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
-// read a bf16x8 fragment from a lane-linear [rows][KB] tile with the
-// (row&7)<<4 byte XOR swizzle (matches the staging source permutation)
+// read a bf16x8 fragment from a lane-linear [rows][KB] tile with a byte
+// XOR swizzle matching the staging source permutation. The XOR spans the
+// row's 16-B columns: 8 columns at KB=64 (<=2-way residual conflict), 16 at
+// KB=128 (conflict-free - rows of a fragment group are distinct mod 16).
+template <int KB>
+constexpr int swz_mask() { return (KB / 8 < 16 ? KB / 8 : 16) - 1; }
+
 template <int KB>  // K extent of the tile in elements
 __device__ __forceinline__ bf16x8 lds_read_swz(const __bf16* tile, int row,
                                                int kelem) {
-    const int byte = (row * (KB * 2) + kelem * 2) ^ ((row & 7) << 4);
+    const int byte = (row * (KB * 2) + kelem * 2) ^
+                     ((row & swz_mask<KB>()) << 4);
     return *reinterpret_cast<const bf16x8*>(
         reinterpret_cast<const char*>(tile) + byte);
 }
@@ -213,7 +219,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         for (int i = 0; i < TBM / (32 / KT); ++i) {
             const int p = wid * (TBM / (32 / KT)) + i;
             const int trow = p * RPP + srow;
-            const int csrc = cdst ^ (trow & 7);
+            const int csrc = cdst ^ (trow & swz_mask<KB>());
             const long arow = min(bm + trow, M - 1);
             glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * RPP][0]);
         }
@@ -221,7 +227,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
         for (int i = 0; i < TBN / (32 / KT); ++i) {
             const int p = wid * (TBN / (32 / KT)) + i;
             const int trow = p * RPP + srow;
-            const int csrc = cdst ^ (trow & 7);
+            const int csrc = cdst ^ (trow & swz_mask<KB>());
             const long brow = min(bn + trow, N - 1);
             glds_b128(B + brow * K + k0 + csrc * 8,
                       &smem[buf][TBM + p * RPP][0]);
