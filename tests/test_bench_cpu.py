@@ -57,3 +57,25 @@ def test_bench_torchrun_two_ranks_cpu():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "pp2"
     assert out["config"]["partition"] == [[1, 24], [25, 48]]
+
+
+def test_bench_torchrun_eight_ranks_cpu():
+    """Rehearsal of the driver's 8-rank launch (gloo on CPU)."""
+    env = dict(os.environ, OMP_NUM_THREADS="2")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29643", os.path.join(REPO, "bench.py"),
+         "--gpus", "8", "--steps", "2", "--warmup", "1",
+         "--batch-size", "8", "--ubatch-size", "4",
+         "--model", "facebook/deit-tiny-distilled-patch16-224"],
+        capture_output=True, timeout=600, env=env, cwd=REPO, check=False)
+    assert proc.returncode == 0, proc.stderr.decode()[-2000:]
+    lines = [l for l in proc.stdout.decode().splitlines()
+             if l.startswith("{")]
+    out = _check_json(lines[-1])
+    assert out["n_gpus"] == 8
+    assert out["config"]["partition"][0] == [1, 6]
+    assert out["config"]["partition"][-1] == [43, 48]

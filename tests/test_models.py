@@ -129,3 +129,13 @@ def test_tuple_activation_boundary_shapes():
     out2 = s2(out)
     assert isinstance(out2, tuple)
     assert out2[0].shape == (4, 197, 3072)  # mlp hidden after sublayer 2
+
+
+def test_vit_huge_head_dim_80_shard():
+    """ViT-Huge: patch 14, head dim 80 (the DHP=96 attention path)."""
+    cfg = get_config("google/vit-huge-patch14-224-in21k")
+    assert cfg.head_dim == 80
+    assert cfg.seq_len == 257
+    shard = vit.ViTShardForImageClassification(cfg, _shard_cfg(1, 4, 128), None)
+    out = shard(torch.randn(1, 3, 224, 224))
+    assert out.shape == (1, 257, 1280)
