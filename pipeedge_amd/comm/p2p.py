@@ -289,6 +289,28 @@ class DistP2pPipelineStage:
             hook, args = hook_entry
             hook(*args, *extra)
 
+    # -- microbatch stream overlap -------------------------------------------
+    def _ustreams(self):
+        """Two HIP streams alternate over consecutive microbatches so their
+        kernels co-reside on the chip (the per-microbatch kernels underfill
+        256 CUs; this is the stream-level analogue of the reference's 3-deep
+        in-flight stage queues, comm/p2p/__init__.py:374-376). Disable with
+        PIPEEDGE_UBATCH_STREAMS=0."""
+        import os
+        if self.device.type != "cuda" or \
+                os.environ.get("PIPEEDGE_UBATCH_STREAMS", "1") == "0":
+            return None
+        if not hasattr(self, "_ustream_pair"):
+            self._ustream_pair = [torch.cuda.Stream(), torch.cuda.Stream()]
+        return self._ustream_pair
+
+    @staticmethod
+    def _stream_ctx(streams, i):
+        import contextlib
+        if streams is None:
+            return contextlib.nullcontext()
+        return torch.cuda.stream(streams[i % 2])
+
     # -- stage-rank batch processing -----------------------------------------
     def stage_run_batch(self, n_ubatch: int):
         """Process one announced batch: exactly ``n_ubatch`` recv->work->send."""
@@ -296,21 +318,26 @@ class DistP2pPipelineStage:
         recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device) \
             if self.rank_src is not None else None
         send = _SendWindow(self.rank_dst) if self.rank_dst is not None else None
+        streams = self._ustreams()
         if recv is not None:
             recv.post(2, n_ubatch)
         for i in range(n_ubatch):
-            self._call_hook(self._recv_pre_hook)
-            wire_in = recv.wait(i)
-            self._call_hook(self._recv_post_hook, wire_in)
-            inputs = self.recv_transform(wire_in)
-            outputs = self.work_cb(inputs)
-            recv.mark_consumed(i)
+            with self._stream_ctx(streams, i):
+                self._call_hook(self._recv_pre_hook)
+                wire_in = recv.wait(i)
+                self._call_hook(self._recv_post_hook, wire_in)
+                inputs = self.recv_transform(wire_in)
+                outputs = self.work_cb(inputs)
+                recv.mark_consumed(i)
+                wire_out = self.send_transform(outputs)
+                self._call_hook(self._send_pre_hook)
+                send.send(wire_out)
+                self._call_hook(self._send_post_hook, wire_out)
             recv.post(i + 3, n_ubatch)
-            wire_out = self.send_transform(outputs)
-            self._call_hook(self._send_pre_hook)
-            send.send(wire_out)
-            self._call_hook(self._send_post_hook, wire_out)
         send.drain()
+        if streams is not None:
+            for s in streams:
+                s.synchronize()
 
     # -- data-rank batch processing ------------------------------------------
     def data_run_batch(self, ubatches: Sequence[ModuleShardData]) -> None:
@@ -320,22 +347,29 @@ class DistP2pPipelineStage:
         n = len(ubatches)
         if self.work_cb is not None and self.rank_dst is None:
             # single-rank degenerate pipeline (reference runtime.py:324-329)
-            for ub in ubatches:
-                self.results_cb(self.work_cb(self._to_device(ub)))
+            streams = self._ustreams()
+            for i, ub in enumerate(ubatches):
+                with self._stream_ctx(streams, i):
+                    self.results_cb(self.work_cb(self._to_device(ub)))
+            if streams is not None:
+                for s in streams:
+                    s.synchronize()
             return
         send = _SendWindow(self.rank_dst)
         recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device)
+        streams = self._ustreams()
         recv.post(2, n)
         collected = 0
         for i, ub in enumerate(ubatches):
-            if self.work_cb is not None:
-                outputs = self.work_cb(self._to_device(ub))
-                wire = self.send_transform(outputs)
-            else:
-                wire = _default_send_transform(self._to_device(ub))
-            self._call_hook(self._send_pre_hook)
-            send.send(wire)
-            self._call_hook(self._send_post_hook, wire)
+            with self._stream_ctx(streams, i):
+                if self.work_cb is not None:
+                    outputs = self.work_cb(self._to_device(ub))
+                    wire = self.send_transform(outputs)
+                else:
+                    wire = _default_send_transform(self._to_device(ub))
+                self._call_hook(self._send_pre_hook)
+                send.send(wire)
+                self._call_hook(self._send_post_hook, wire)
             # opportunistically drain finished results to bound buffering
             while (collected < i and collected in recv.works and
                    all(w.is_completed() for w in recv.works[collected])):

@@ -192,7 +192,8 @@ void launch_ln(const __bf16* x, const __bf16* res, const __bf16* w,
     auto stream = pe_current_stream();
     static const bool split = []() {
         const char* e = std::getenv("PIPEEDGE_LN_SPLIT");
-        return e == nullptr || std::atoi(e) != 0;  // default on (HW sweep)
+        return e != nullptr && std::atoi(e) != 0;  // default OFF (HW sweep:
+        // the 1-wave register path wins at transformer row counts)
     }();
     if (split && D % 16 == 0 && D / 2 <= 1024) {
         // 2 waves per row, 2 rows per block

@@ -41,9 +41,12 @@ class GraphedModule:
 
     @staticmethod
     def _key(data):
+        # keyed by stream too: concurrent replays on different streams must
+        # not share capture-static buffers
+        sid = torch.cuda.current_stream().cuda_stream
         if isinstance(data, torch.Tensor):
-            return (tuple(data.shape), data.dtype)
-        return tuple((tuple(t.shape), t.dtype) for t in data)
+            return (sid, tuple(data.shape), data.dtype)
+        return (sid,) + tuple((tuple(t.shape), t.dtype) for t in data)
 
     def _capture(self, data):
         static_in = data.clone() if isinstance(data, torch.Tensor) \
