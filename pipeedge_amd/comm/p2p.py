@@ -297,11 +297,11 @@ class DistP2pPipelineStage:
         in-flight stage queues, comm/p2p/__init__.py:374-376). Disable with
         PIPEEDGE_UBATCH_STREAMS=0."""
         import os
-        if self.device.type != "cuda" or \
-                os.environ.get("PIPEEDGE_UBATCH_STREAMS", "1") == "0":
+        n = int(os.environ.get("PIPEEDGE_UBATCH_STREAMS", "2"))
+        if self.device.type != "cuda" or n < 2:
             return None
         if not hasattr(self, "_ustream_pair"):
-            self._ustream_pair = [torch.cuda.Stream(), torch.cuda.Stream()]
+            self._ustream_pair = [torch.cuda.Stream() for _ in range(n)]
         return self._ustream_pair
 
     @staticmethod
@@ -309,18 +309,20 @@ class DistP2pPipelineStage:
         import contextlib
         if streams is None:
             return contextlib.nullcontext()
-        return torch.cuda.stream(streams[i % 2])
+        return torch.cuda.stream(streams[i % len(streams)])
 
     # -- stage-rank batch processing -----------------------------------------
     def stage_run_batch(self, n_ubatch: int):
         """Process one announced batch: exactly ``n_ubatch`` recv->work->send."""
         assert self.work_cb is not None
-        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device) \
+        streams = self._ustreams()
+        depth = max(2, len(streams) if streams else 2)
+        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device,
+                             depth=depth) \
             if self.rank_src is not None else None
         send = _SendWindow(self.rank_dst) if self.rank_dst is not None else None
-        streams = self._ustreams()
         if recv is not None:
-            recv.post(2, n_ubatch)
+            recv.post(depth, n_ubatch)
         for i in range(n_ubatch):
             with self._stream_ctx(streams, i):
                 self._call_hook(self._recv_pre_hook)
@@ -333,7 +335,7 @@ class DistP2pPipelineStage:
                 self._call_hook(self._send_pre_hook)
                 send.send(wire_out)
                 self._call_hook(self._send_post_hook, wire_out)
-            recv.post(i + 3, n_ubatch)
+            recv.post(i + depth + 1, n_ubatch)
         send.drain()
         if streams is not None:
             for s in streams:
