@@ -91,3 +91,52 @@ def test_runtime_auto_sched_from_yaml(tmp_path):
     assert "throughput is" in outs[0][1]
     # monitoring CSVs were written by the run
     assert (tmp_path / "shard.csv").exists()
+
+
+def _run_n(port, n, extra_args, env_extra=None):
+    ctx = mp.get_context("spawn")
+    result_q = ctx.Queue()
+    procs = []
+    for r in range(n):
+        p = ctx.Process(target=_worker_n,
+                        args=(r, n, port, extra_args, env_extra or {},
+                              result_q))
+        p.start()
+        procs.append(p)
+    outs = {}
+    for _ in range(n):
+        rank, rc, out, err = result_q.get(timeout=300)
+        outs[rank] = (rc, out, err)
+    for p in procs:
+        p.join(timeout=30)
+    return outs
+
+
+def _worker_n(rank, world, port, extra_args, env_extra, result_q):
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+               PYTHONPATH=REPO, OMP_NUM_THREADS="2", **env_extra)
+    cmd = [sys.executable, os.path.join(REPO, "runtime.py"), str(rank),
+           str(world), "-d", "cpu", "--port", str(port)] + extra_args
+    proc = subprocess.run(cmd, capture_output=True, timeout=300, env=env,
+                          cwd=REPO, check=False)
+    result_q.put((rank, proc.returncode, proc.stdout.decode(),
+                  proc.stderr.decode()[-1200:]))
+
+
+def test_runtime_pure_data_rank():
+    """Data rank that is NOT a pipeline stage (reference model_cfg.py:133-141
+    wiring): rank 2 feeds inputs / collects results; ranks 0-1 compute."""
+    outs = _run_n(29637, 3, ["-b", "4", "-u", "2", "-m", MODEL,
+                             "-pt", "1,24,25,48", "-r", "0,1", "-D", "2"])
+    for r in range(3):
+        assert outs[r][0] == 0, (r, outs[r][2])
+    assert "throughput is" in outs[2][1]
+
+
+def test_runtime_reversed_rank_order():
+    """Stage order reversed relative to ranks (-r 1,0 with data on rank 1)."""
+    outs = _run_n(29639, 2, ["-b", "4", "-u", "2", "-m", MODEL,
+                             "-pt", "1,24,25,48", "-r", "1,0", "-D", "1"])
+    for r in range(2):
+        assert outs[r][0] == 0, (r, outs[r][2])
+    assert "throughput is" in outs[1][1]
