@@ -16,20 +16,39 @@ from typing import Dict, List, Optional
 
 
 class EnergySource:
-    """Total-energy reader in microjoules; prefers AMD GPU hwmon counters."""
+    """Energy metering from hwmon (the EnergyMon equivalent - SURVEY.md 5.5).
+
+    Prefers cumulative ``energy*_input`` counters (microjoules); falls back
+    to ``power*_input`` (microwatts, what MI355X boxes expose), which the
+    monitor integrates over each iteration (trapezoid)."""
 
     def __init__(self):
         self._files = []
+        self.kind = "none"
         for path in sorted(glob.glob("/sys/class/hwmon/hwmon*/energy*_input")):
-            try:
-                with open(path) as f:
-                    int(f.read().strip())
+            if self._probe(path):
                 self._files.append(path)
-            except (OSError, ValueError):
-                continue
-        self.name = "hwmon" if self._files else "none"
+        if self._files:
+            self.kind = "energy"
+        else:
+            for path in sorted(
+                    glob.glob("/sys/class/hwmon/hwmon*/power*_input")):
+                if self._probe(path):
+                    self._files.append(path)
+            if self._files:
+                self.kind = "power"
+        self.name = "hwmon-" + self.kind if self._files else "none"
 
-    def read_uj(self) -> int:
+    @staticmethod
+    def _probe(path) -> bool:
+        try:
+            with open(path) as f:
+                int(f.read().strip())
+            return True
+        except (OSError, ValueError):
+            return False
+
+    def _read_sum(self) -> int:
         total = 0
         for path in self._files:
             try:
@@ -38,6 +57,14 @@ class EnergySource:
             except (OSError, ValueError):
                 pass
         return total
+
+    def read_uj(self) -> int:
+        """Cumulative energy in microjoules (energy-counter sources only)."""
+        return self._read_sum() if self.kind == "energy" else 0
+
+    def read_uw(self) -> int:
+        """Instantaneous power in microwatts (power sources only)."""
+        return self._read_sum() if self.kind == "power" else 0
 
 
 class _HeartbeatRecord:
@@ -129,15 +156,21 @@ class Heartbeat:
 
 
 class MonitorIterationContext:
-    """Tracks one in-flight iteration (start time + start energy)."""
+    """Tracks one in-flight iteration (start time + start energy/power)."""
 
     def __init__(self):
         self.t_start = None
         self.e_start = 0
+        self.p_start = 0
 
     def start(self, energy_source: Optional[EnergySource]):
         self.t_start = time.perf_counter()
-        self.e_start = energy_source.read_uj() if energy_source else 0
+        if energy_source is not None:
+            self.e_start = energy_source.read_uj()
+            self.p_start = energy_source.read_uw()
+        else:
+            self.e_start = 0
+            self.p_start = 0
 
 
 class MonitorContext:
@@ -187,8 +220,14 @@ class MonitorContext:
             ctx = self._iter_ctx.setdefault(key, MonitorIterationContext())
         now = time.perf_counter()
         t_start = ctx.t_start if ctx.t_start is not None else now
-        energy = (self.energy_source.read_uj() - ctx.e_start) \
-            if (self.energy_source and ctx.t_start is not None) else 0
+        energy = 0
+        if self.energy_source is not None and ctx.t_start is not None:
+            if self.energy_source.kind == "energy":
+                energy = self.energy_source.read_uj() - ctx.e_start
+            elif self.energy_source.kind == "power":
+                # trapezoid integration of the power samples: uW * s = uJ
+                p_end = self.energy_source.read_uw()
+                energy = int(0.5 * (ctx.p_start + p_end) * (now - t_start))
         hb = self._hbs[key]
         tag = self._tags[key]
         self._tags[key] = tag + 1
