@@ -234,6 +234,186 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Small-sequence variant (S <= 256, dh <= 64): the WHOLE K/V fits in LDS, so
+// there is exactly ONE staging barrier per block and no online softmax -
+// each wave holds its 16 q-rows' full score row (ceil(S/16) fragments) in
+// registers, does a full-row max/sum (DPP within the 16-lane groups), then
+// streams P->PV in wave-private chunks with zero further synchronization.
+// This is the ViT/DeiT shape (S = 197/198).
+// ---------------------------------------------------------------------------
+template <int DHP, int SMAX>
+__global__ __launch_bounds__(256) void qkv_attention_smallS_kernel(
+        const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
+        int B, int S, int H, int dh, float scale) {
+    constexpr int NF_MAX = SMAX / 16;
+    __shared__ __bf16 Ks[SMAX][DHP + PAD];
+    __shared__ __bf16 Vs[DHP][SMAX + PAD];   // transposed, XOR-swizzled
+    __shared__ __bf16 Ps[4][16][64 + PAD];   // per-wave P chunk (64 kv)
+
+    auto vs_at = [&](int d, int s) -> __bf16* {
+        int byte = (d * (SMAX + PAD) + s) * 2;
+        byte ^= ((d >> 3) & 7) << 4;
+        return reinterpret_cast<__bf16*>(
+            reinterpret_cast<char*>(&Vs[0][0]) + byte);
+    };
+    auto ps_at = [&](int w, int row, int col) -> __bf16* {
+        int byte = ((w * 16 + row) * (64 + PAD) + col) * 2;
+        byte ^= ((row >> 3) & 1) << 5;
+        return reinterpret_cast<__bf16*>(
+            reinterpret_cast<char*>(&Ps[0][0][0]) + byte);
+    };
+
+    const int bh = blockIdx.x;
+    const int b = bh / H;
+    const int h = bh % H;
+    const int q0 = blockIdx.y * QCH;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int D = H * dh;
+    const long row_stride = 3L * D;
+    const __bf16* q_base = qkv + (long)b * S * row_stride + h * dh;
+    const __bf16* k_base = q_base + D;
+    const __bf16* v_base = q_base + 2L * D;
+
+    // ---- stage the whole K (natural) and V (transposed image) ----
+    // stage (and zero) up to the PV chunk boundary: the last 64-kv chunk
+    // reads V rows past S, which must be zeros, not stale LDS (0 * NaN)
+    const int sp = (S + 15) & ~15;        // score-fragment extent
+    const int sp_chunk = (S + 63) & ~63;  // PV chunk extent
+    for (int piece = tid; piece < SMAX * (DHP / 8); piece += 256) {
+        const int r = piece / (DHP / 8);
+        const int c = (piece % (DHP / 8)) * 8;
+        if (r >= sp_chunk) break;
+        bf16x8 kv_ = {};
+        bf16x8 vv = {};
+        if (r < S && c < dh) {
+            kv_ = *reinterpret_cast<const bf16x8*>(
+                k_base + (long)r * row_stride + c);
+            vv = *reinterpret_cast<const bf16x8*>(
+                v_base + (long)r * row_stride + c);
+        }
+        *reinterpret_cast<bf16x8*>(&Ks[r][c]) = kv_;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) *vs_at(c + j, r) = vv[j];
+    }
+    __syncthreads();
+
+    // ---- per-wave: 16 q rows, full score row in registers ----
+    const int qrow = q0 + wid * 16 + (lane & 15);
+    bf16x8 qfrag[DHP / 32];
+    #pragma unroll
+    for (int kk = 0; kk < DHP; kk += 32) {
+        const int kb = kk + (lane >> 4) * 8;
+        bf16x8 v = {};
+        if (qrow < S && kb < dh)
+            v = *reinterpret_cast<const bf16x8*>(
+                q_base + (long)qrow * row_stride + kb);
+        qfrag[kk / 32] = v;
+    }
+
+    // sfrag must ONLY be indexed with compile-time constants (a runtime
+    // index sends the whole array to scratch - guide rule 20), so every
+    // nf loop is fully unrolled over NF_MAX with a runtime guard.
+    const int nf_count = sp / 16;
+    f32x4 sfrag[NF_MAX];
+    __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int nf = 0; nf < NF_MAX; ++nf) {
+        if (nf >= nf_count) continue;
+        f32x4 acc = {};
+        #pragma unroll
+        for (int kk = 0; kk < DHP; kk += 32) {
+            const int kb = kk + (lane >> 4) * 8;
+            const bf16x8 bb = *reinterpret_cast<const bf16x8*>(
+                &Ks[nf * 16 + (lane & 15)][kb]);
+            acc = mfma_16x16x32_bf16(qfrag[kk / 32], bb, acc);
+        }
+        sfrag[nf] = acc;
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // scale + mask + full-row softmax (no rescale pass: the row is whole)
+    #pragma unroll
+    for (int nf = 0; nf < NF_MAX; ++nf) {
+        if (nf >= nf_count) continue;
+        const bool valid = nf * 16 + (lane & 15) < S;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+            sfrag[nf][r] = valid ? sfrag[nf][r] * scale : -1e30f;
+    }
+    float mrow[4], lrow[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        float mt = -1e30f;
+        #pragma unroll
+        for (int nf = 0; nf < NF_MAX; ++nf)
+            if (nf < nf_count)
+                mt = fmaxf(mt, sfrag[nf][r]);
+        mrow[r] = group16_reduce(mt, MaxOp());
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        float s = 0.0f;
+        #pragma unroll
+        for (int nf = 0; nf < NF_MAX; ++nf) {
+            if (nf >= nf_count) continue;
+            const float p = __expf(sfrag[nf][r] - mrow[r]);
+            sfrag[nf][r] = p;
+            s += p;
+        }
+        lrow[r] = group16_reduce(s, SumOp());
+    }
+
+    // ---- PV in wave-private 64-kv chunks (no barriers: K/V read-only,
+    // Ps chunk is this wave's own) ----
+    f32x4 o_acc[DHP / 16] = {};
+    #pragma unroll
+    for (int c0 = 0; c0 < NF_MAX; c0 += 4) {
+        if (c0 >= nf_count) continue;
+        #pragma unroll
+        for (int cf = 0; cf < 4; ++cf) {
+            const f32x4 pv = (c0 + cf < NF_MAX)
+                ? sfrag[c0 + cf < NF_MAX ? c0 + cf : 0] : f32x4{};
+            const bool live = c0 + cf < nf_count;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                *ps_at(wid, (lane >> 4) * 4 + r, cf * 16 + (lane & 15)) =
+                    f2bf(live ? pv[r] : 0.0f);
+        }
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int kk = 0; kk < 64; kk += 32) {
+            const int kb = kk + (lane >> 4) * 8;
+            const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                ps_at(wid, lane & 15, kb));
+            #pragma unroll
+            for (int nfo = 0; nfo < DHP / 16; ++nfo) {
+                const bf16x8 bb = *reinterpret_cast<const bf16x8*>(
+                    vs_at(nfo * 16 + (lane & 15), c0 * 16 + kb));
+                o_acc[nfo] = mfma_16x16x32_bf16(a, bb, o_acc[nfo]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- epilogue ----
+    #pragma unroll
+    for (int nfo = 0; nfo < DHP / 16; ++nfo) {
+        const int d = nfo * 16 + (lane & 15);
+        if (d >= dh) continue;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int srow = q0 + wid * 16 + (lane >> 4) * 4 + r;
+            if (srow >= S) continue;
+            const float denom = lrow[r] > 0.0f ? lrow[r] : 1.0f;
+            ctx[((long)b * S + srow) * D + h * dh + d] =
+                f2bf(o_acc[nfo][r] / denom);
+        }
+    }
+}
+
 }  // namespace
 
 torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
@@ -256,7 +436,19 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         const char* e = std::getenv("PIPEEDGE_ATTN_KVT");
         return e ? std::atoi(e) : 128;  // on-HW sweep: 128 wins for S >= 128
     }();
-    if (dh <= 64) {
+    static const bool small_s = []() {
+        // default OFF: on-HW A/B measured the whole-KV single-barrier
+        // variant at 13.5us vs 11.5us for the tiled kernel at S=197 (its
+        // 153-VGPR score-row residency costs more occupancy than the
+        // barrier removal buys); enable to experiment.
+        const char* e = std::getenv("PIPEEDGE_ATTN_SMALLS");
+        return e != nullptr && std::atoi(e) != 0;
+    }();
+    if (dh <= 64 && S <= 256 && small_s) {
+        hipLaunchKernelGGL((qkv_attention_smallS_kernel<64, 256>), grid,
+                           dim3(256), 0, pe_current_stream(), in, out,
+                           B, S, H, dh, (float)scale);
+    } else if (dh <= 64) {
         if (kvt >= 128 && S >= 128) {
             hipLaunchKernelGGL((qkv_attention_kernel<64, 128>), grid,
                                dim3(256), 0, pe_current_stream(), in, out,
