@@ -60,9 +60,13 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     dtype = torch.bfloat16 if use_cuda else torch.float32
 
     distributed = world_size > 1
+    results_group = None
     if distributed:
         backend = "nccl" if use_cuda else "gloo"
         dist.init_process_group(backend, rank=rank, world_size=world_size)
+        # dedicated communicator for the results back-edge (see
+        # DistP2pPipelineStage group docs)
+        results_group = dist.new_group(backend=backend)
 
     model_name = args.model
     total_layers = model_cfg.get_model_layers(model_name)
@@ -82,7 +86,8 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     stage_ctx = model_cfg.dist_p2p_pipeline_stage_factory(
         stage_ranks, data_rank, rank, stage, work_cb, results.append,
         model_name=model_name, ubatch_size=args.ubatch_size,
-        stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype)
+        stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype,
+        results_group=results_group)
 
     # synthetic data, resident on-device before the timed region
     config = model_cfg.get_model_config(model_name)

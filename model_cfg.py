@@ -168,13 +168,15 @@ def dist_p2p_pipeline_stage_factory(stage_ranks: List[int], data_rank: int,
                                     ubatch_size: int = 8,
                                     stage_layers: List[Tuple[int, int]] = None,
                                     stage_quant: List[int] = None,
-                                    dtype: torch.dtype = torch.float32) \
+                                    dtype: torch.dtype = torch.float32,
+                                    results_group=None) \
         -> p2p.DistP2pPipelineStage:
     """Wire this rank's role in the p2p pipeline (reference model_cfg.py:128-166)
     and attach the static recv/send schemas for its edges."""
     edges = stage_wire_schemas(model_name, stage_layers, ubatch_size,
                                stage_quant or [0] * len(stage_layers), dtype)
     num_stages = len(stage_ranks)
+    recv_group = send_group = None
     if rank == data_rank:
         if stage is None:
             rank_src = stage_ranks[-1]
@@ -190,6 +192,7 @@ def dist_p2p_pipeline_stage_factory(stage_ranks: List[int], data_rank: int,
         results_cb = handle_results_cb
         recv_specs = edges[-1]          # results edge
         send_specs = edges[1] if work_cb is not None else edges[0]
+        recv_group = results_group      # back-edge: its own communicator
     elif stage is None:
         return p2p.DistP2pPipelineStage(None, None, [], [], None, None)
     else:
@@ -199,6 +202,10 @@ def dist_p2p_pipeline_stage_factory(stage_ranks: List[int], data_rank: int,
         results_cb = None
         recv_specs = edges[stage]
         send_specs = edges[stage + 1]
+        if stage == num_stages - 1 and rank_dst == data_rank:
+            send_group = results_group  # back-edge sender side
     return p2p.DistP2pPipelineStage(rank_src, rank_dst, recv_specs, send_specs,
                                     work_cb, results_cb,
-                                    device=devices.DEVICE)
+                                    device=devices.DEVICE,
+                                    recv_group=recv_group,
+                                    send_group=send_group)

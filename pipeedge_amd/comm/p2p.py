@@ -118,10 +118,10 @@ class DistP2pContext(DistContext):
         dist.init_process_group(self._backend, rank=self._rank,
                                 world_size=self._world_size)
         # control plane always on gloo (CPU tensors, any-source recv, tags)
-        if self._backend == "gloo":
-            self._ctrl_group = dist.new_group(backend="gloo")
-        else:
-            self._ctrl_group = dist.new_group(backend="gloo")
+        self._ctrl_group = dist.new_group(backend="gloo")
+        # dedicated communicator for the results back-edge (see
+        # DistP2pPipelineStage)
+        self.results_group = dist.new_group(backend=self._backend)
         self._cmd_thread = CommandThread(self._ctrl_group, self._cmd_cb,
                                          self.stop_event)
         self._cmd_thread.start()
@@ -171,8 +171,9 @@ class _RecvPipeline:
     """Double-buffered in-order irecv stream from one peer."""
 
     def __init__(self, src: int, specs: Sequence[TensorSpec], device,
-                 depth: int = 2):
+                 depth: int = 2, group=None):
         self.src = src
+        self.group = group
         self.device = torch.device(device)
         self.specs = list(specs)
         self.depth = depth
@@ -191,7 +192,8 @@ class _RecvPipeline:
             if self.events[s] is not None:
                 self.events[s].synchronize()
                 self.events[s] = None
-            self.works[i] = [dist.irecv(t, src=self.src, tag=TAG_DATA)
+            self.works[i] = [dist.irecv(t, src=self.src, group=self.group,
+                                        tag=TAG_DATA)
                              for t in self.slots[s]]
             self.next_post += 1
 
@@ -212,8 +214,9 @@ class _RecvPipeline:
 class _SendWindow:
     """Outstanding isend tracker: keeps tensors alive, bounds in-flight count."""
 
-    def __init__(self, dst: int, max_inflight: int = 8):
+    def __init__(self, dst: int, max_inflight: int = 8, group=None):
         self.dst = dst
+        self.group = group
         self.max_inflight = max_inflight
         self.pending = collections.deque()
 
@@ -223,7 +226,8 @@ class _SendWindow:
             for w in works:
                 w.wait()
         tensors = [t.contiguous() for t in tensors]
-        works = [dist.isend(t, dst=self.dst, tag=TAG_DATA) for t in tensors]
+        works = [dist.isend(t, dst=self.dst, group=self.group, tag=TAG_DATA)
+                 for t in tensors]
         self.pending.append((works, tensors))
 
     def drain(self):
@@ -251,9 +255,17 @@ class DistP2pPipelineStage:
     def __init__(self, rank_src: Optional[int], rank_dst: Optional[int],
                  recv_specs: Sequence[TensorSpec], send_specs: Sequence[TensorSpec],
                  work_cb: Optional[Callable], results_cb: Optional[Callable],
-                 device=None):
+                 device=None, recv_group=None, send_group=None):
+        # Separate process groups break the pipeline's only CYCLE (the
+        # results back-edge last-stage -> data-rank): with a single NCCL
+        # communicator, the data rank's pre-posted result irecvs could
+        # serialize ahead of its input isends on the comm stream and
+        # deadlock the ring. The forward chain is acyclic and safe on one
+        # group; the back-edge gets its own communicator.
         self.rank_src = rank_src
         self.rank_dst = rank_dst
+        self.recv_group = recv_group
+        self.send_group = send_group
         self.recv_specs = list(recv_specs) if recv_specs else []
         self.send_specs = list(send_specs) if send_specs else []
         self.work_cb = work_cb
@@ -318,9 +330,10 @@ class DistP2pPipelineStage:
         streams = self._ustreams()
         depth = max(2, len(streams) if streams else 2)
         recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device,
-                             depth=depth) \
+                             depth=depth, group=self.recv_group) \
             if self.rank_src is not None else None
-        send = _SendWindow(self.rank_dst) if self.rank_dst is not None else None
+        send = _SendWindow(self.rank_dst, group=self.send_group) \
+            if self.rank_dst is not None else None
         if recv is not None:
             recv.post(depth, n_ubatch)
         for i in range(n_ubatch):
@@ -357,8 +370,9 @@ class DistP2pPipelineStage:
                 for s in streams:
                     s.synchronize()
             return
-        send = _SendWindow(self.rank_dst)
-        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device)
+        send = _SendWindow(self.rank_dst, group=self.send_group)
+        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device,
+                             group=self.recv_group)
         streams = self._ustreams()
         recv.post(2, n)
         collected = 0
