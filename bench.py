@@ -87,11 +87,12 @@ def run_rank(args, rank: int, world_size: int) -> dict:
         stage_ranks, data_rank, rank, stage, work_cb, results.append,
         model_name=model_name, ubatch_size=args.ubatch_size,
         stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype,
-        results_group=results_group)
+        results_group=results_group, seq_len=args.seq_len)
 
     # synthetic data, resident on-device before the timed region
     config = model_cfg.get_model_config(model_name)
-    dataset = load_dataset_synthetic(config, args.batch_size)
+    dataset = load_dataset_synthetic(config, args.batch_size,
+                                     seq_len=args.seq_len)
     n_ubatch = args.batch_size // args.ubatch_size
     ubatches = []
     if rank == data_rank:
@@ -160,7 +161,9 @@ def run_rank(args, rank: int, world_size: int) -> dict:
                 "model": model_name,
                 "global_batch": args.batch_size,
                 "ubatch_size": args.ubatch_size,
-                "seq_len": 197,
+                "seq_len": args.seq_len or
+                (config.seq_len if config.model_type in ("vit", "deit")
+                 else config.max_position_embeddings),
                 "parallelism": f"pp{world_size}",
                 "partition": [list(p) for p in stage_layers],
                 "weights": "random-init",
@@ -183,6 +186,8 @@ def main():
     parser.add_argument("--model", type=str,
                         default="google/vit-base-patch16-224")
     parser.add_argument("--model-file", type=str, default=None)
+    parser.add_argument("--seq-len", type=int, default=None,
+                        help="token seq length for text models (BERT)")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))

@@ -139,7 +139,7 @@ def dist_rpc_pipeline_factory(model_name: str, model_file: Optional[str],
 
 def stage_wire_schemas(model_name: str, stage_layers: List[Tuple[int, int]],
                        ubatch_size: int, stage_quant: List[int],
-                       dtype: torch.dtype):
+                       dtype: torch.dtype, seq_len: Optional[int] = None):
     """Compute every pipeline edge's wire schema.
 
     Edges: data->stage0 (model input), stage i -> stage i+1 (boundary,
@@ -150,9 +150,10 @@ def stage_wire_schemas(model_name: str, stage_layers: List[Tuple[int, int]],
     """
     config = get_model_config(model_name)
     total = get_model_layers(model_name)
-    edges = [spec.input_spec(config, ubatch_size, dtype)]
+    edges = [spec.input_spec(config, ubatch_size, dtype, seq_len)]
     for i, (_, layer_end) in enumerate(stage_layers):
-        base = spec.stage_boundary_spec(config, layer_end, total, ubatch_size, dtype)
+        base = spec.stage_boundary_spec(config, layer_end, total, ubatch_size,
+                                        dtype, seq_len)
         if layer_end < total:
             qbit = stage_quant[i] if stage_quant else 0
             base = spec.quant_wire_spec(base, qbit)
@@ -169,12 +170,14 @@ def dist_p2p_pipeline_stage_factory(stage_ranks: List[int], data_rank: int,
                                     stage_layers: List[Tuple[int, int]] = None,
                                     stage_quant: List[int] = None,
                                     dtype: torch.dtype = torch.float32,
-                                    results_group=None) \
+                                    results_group=None,
+                                    seq_len: Optional[int] = None) \
         -> p2p.DistP2pPipelineStage:
     """Wire this rank's role in the p2p pipeline (reference model_cfg.py:128-166)
     and attach the static recv/send schemas for its edges."""
     edges = stage_wire_schemas(model_name, stage_layers, ubatch_size,
-                               stage_quant or [0] * len(stage_layers), dtype)
+                               stage_quant or [0] * len(stage_layers), dtype,
+                               seq_len)
     num_stages = len(stage_ranks)
     recv_group = send_group = None
     if rank == data_rank:

@@ -21,13 +21,14 @@ from .config import TransformerConfig
 
 
 def input_spec(config: TransformerConfig, ubatch_size: int,
-               dtype: torch.dtype) -> List[TensorSpec]:
+               dtype: torch.dtype,
+               seq_len: Optional[int] = None) -> List[TensorSpec]:
     """Schema of the model input (what the data rank sends to stage 0)."""
     if config.model_type in ("vit", "deit"):
         return [TensorSpec((ubatch_size, config.num_channels, config.image_size,
                             config.image_size), dtype)]
-    # BERT: token ids, fixed benchmark seq of max positions unless overridden
-    return [TensorSpec((ubatch_size, config.max_position_embeddings), torch.int64)]
+    s = seq_len if seq_len else config.max_position_embeddings
+    return [TensorSpec((ubatch_size, s), torch.int64)]
 
 
 def boundary_spec(config: TransformerConfig, layer_end: int, ubatch_size: int,

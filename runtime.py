@@ -484,7 +484,7 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
             stage_ranks, data_rank, rank, stage, work_cb, handle_results,
             model_name=model_name, ubatch_size=ubatch_size,
             stage_layers=stage_layers, stage_quant=stage_quant, dtype=dtype,
-            results_group=dist_ctx.results_group)
+            results_group=dist_ctx.results_group, seq_len=seq_len or None)
 
         # attach quant / cast transforms to the stage's edges
         config = model_cfg.get_model_config(model_name)

@@ -79,3 +79,24 @@ def test_bench_torchrun_eight_ranks_cpu():
     assert out["n_gpus"] == 8
     assert out["config"]["partition"][0] == [1, 6]
     assert out["config"]["partition"][-1] == [43, 48]
+
+
+def test_bench_bert_four_stage_cpu():
+    """BASELINE config #4 shape: BERT-CoLA 4-stage, seq_len=128."""
+    env = dict(os.environ, OMP_NUM_THREADS="2")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29645", os.path.join(REPO, "bench.py"),
+         "--gpus", "4", "--steps", "2", "--warmup", "1",
+         "--batch-size", "8", "--ubatch-size", "4", "--seq-len", "128",
+         "--model", "textattack/bert-base-uncased-CoLA"],
+        capture_output=True, timeout=600, env=env, cwd=REPO, check=False)
+    assert proc.returncode == 0, proc.stderr.decode()[-2000:]
+    lines = [l for l in proc.stdout.decode().splitlines()
+             if l.startswith("{")]
+    out = json.loads(lines[-1])
+    assert out["n_gpus"] == 4
+    assert out["config"]["seq_len"] == 128
