@@ -272,6 +272,110 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// 3-buffer glds pipeline (K % 64 == 0): two K-tiles of loads stay in flight
+// ACROSS barriers (guide "Pipelining across barriers"): raw s_barrier +
+// counted `s_waitcnt vmcnt(N)` instead of __syncthreads(), whose fence
+// would drain the in-flight LDS-DMA to vmcnt(0) every tile. 48 KiB LDS ->
+// 3 blocks/CU. Targets the low-occupancy long-K shapes (fc2 class).
+// ---------------------------------------------------------------------------
+template <int MI, int NI, bool XCD = false, int EPM = 0>
+__global__ __launch_bounds__(256) void gemm_bf16_glds3_kernel(
+        const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+        __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
+    constexpr int TBM = 32 * MI;
+    constexpr int TBN = 32 * NI;
+    constexpr int KB = BK;
+    __shared__ __bf16 smem[3][TBM + TBN][KB];
+
+    int bx = blockIdx.x, by = blockIdx.y;
+    if (XCD) {
+        const int nb = gridDim.x * gridDim.y;
+        const int id = xcd_swizzle(by * gridDim.x + bx, nb);
+        bx = id % gridDim.x;
+        by = id / gridDim.x;
+    }
+    const int bm = bx * TBM;
+    const int bn = by * TBN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wm = (wid >> 1) * (16 * MI);
+    const int wn = (wid & 1) * (16 * NI);
+
+    const int srow = lane >> 3;
+    const int cdst = lane & 7;
+
+    // glds per wave per tile: TBM/32 + TBN/32
+    constexpr int GLDS_PER_TILE = TBM / 32 + TBN / 32;
+    auto stage = [&](int buf, int k0) {
+        #pragma unroll
+        for (int i = 0; i < TBM / 32; ++i) {
+            const int p = wid * (TBM / 32) + i;
+            const int trow = p * 8 + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long arow = min(bm + trow, M - 1);
+            glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * 8][0]);
+        }
+        #pragma unroll
+        for (int i = 0; i < TBN / 32; ++i) {
+            const int p = wid * (TBN / 32) + i;
+            const int trow = p * 8 + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long brow = min(bn + trow, N - 1);
+            glds_b128(B + brow * K + k0 + csrc * 8,
+                      &smem[buf][TBM + p * 8][0]);
+        }
+    };
+
+    f32x4 acc[MI][NI] = {};
+    const int n_tiles = K / KB;
+    stage(0, 0);
+    if (n_tiles > 1) stage(1, KB);
+
+    for (int t = 0; t < n_tiles; ++t) {
+        // tile t's LDS-DMA retired for THIS wave; up to one more tile stays
+        // in flight across the barrier
+        if (t + 1 < n_tiles)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GLDS_PER_TILE)
+                         : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (t + 2 < n_tiles) stage((t + 2) % 3, (t + 2) * KB);
+        const int buf = t % 3;
+        #pragma unroll
+        for (int kk = 0; kk < KB; kk += 32) {
+            const int kgrp = kk + (lane >> 4) * 8;
+            bf16x8 af[MI], bfr[NI];
+            #pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+                af[mi] = lds_read_swz<KB>(&smem[buf][0][0],
+                                          wm + mi * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+                bfr[ni] = lds_read_swz<KB>(&smem[buf][TBM][0],
+                                           wn + ni * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < NI; ++ni)
+                    acc[mi][ni] = mfma_16x16x32_bf16(af[mi], bfr[ni],
+                                                     acc[mi][ni]);
+        }
+        // no trailing barrier: the next iteration's entry barrier orders
+        // every wave's reads of buf before anyone overwrites it (the only
+        // writer of buf t+3 issues after that barrier)
+    }
+    __builtin_amdgcn_s_barrier();
+
+    float* lnbuf = reinterpret_cast<float*>(&smem[0][0][0]);
+    if constexpr (EPM == 1)
+        ln_coeffs_to_lds<TBM>(ep, bm, tid, lnbuf);
+    gemm_epilogue<MI, NI, EPM>(acc, ep, bm, bn, wm, wn, lane,
+                               (bn / TBN) * 2 + (wid & 1), lnbuf);
+}
+
+// ---------------------------------------------------------------------------
 // generic path: any K (padded LDS, plain vectorized staging)
 // ---------------------------------------------------------------------------
 template <int EPM = 0>
@@ -404,15 +508,17 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
         // the `tile` arg (0 = auto heuristic)
         t = (int)tile;
         if (t == 0) {
-            // from the on-HW sweeps (tools/gemm_sweep.py,
-            // tools/hotpath_sweep.py): XCD-swizzled 64x64 default; BK=128
-            // for long-K (fc2 class, +27%); 128x64 for wide-N (fc1 class)
-            if (K >= 2048 && K % (2 * BK) == 0) t = 6;
-            else if (N >= 3072 && M >= 1024) t = 2;
+            // from the on-HW sweeps (tools/gemm_sweep.py, hotpath_sweep.py,
+            // glds3_screen.py): XCD-swizzled 64x64 2-buffer default; the
+            // 3-buffer raw-barrier pipeline (t7) for long-K / wide-N where
+            // per-block serial latency dominates (fc1 18.1->17.0us,
+            // fc2 22.9->22.5us)
+            if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
             else t = 5;
         }
         if (t == 4 && K % (2 * BK) != 0) t = 1;
         if (t == 6 && K % (2 * BK) != 0) t = 5;
+
     }
     torch::Tensor stats;
     if (want_stats) {
@@ -443,7 +549,21 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
     } while (0)
     if (K % BK == 0) {
-        if (t == 4 || t == 5 || t == 6) {
+        if (t == 7) {
+            dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+            if (epm == 1)
+                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 1>),
+                                   grid, dim3(256), 0, pe_current_stream(),
+                                   ap, bp, cp, M, N, K, ep);
+            else if (epm == 2)
+                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 2>),
+                                   grid, dim3(256), 0, pe_current_stream(),
+                                   ap, bp, cp, M, N, K, ep);
+            else
+                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 0>),
+                                   grid, dim3(256), 0, pe_current_stream(),
+                                   ap, bp, cp, M, N, K, ep);
+        } else if (t == 4 || t == 5 || t == 6) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             if (t == 4) PE_LAUNCH_GLDS(2, 2, 2, false, grid);
             else if (t == 5) PE_LAUNCH_GLDS(2, 2, 1, true, grid);
