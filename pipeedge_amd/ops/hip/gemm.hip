@@ -17,6 +17,8 @@
 // (reference vit.py:45-53).
 #include <torch/extension.h>
 
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
@@ -507,6 +509,15 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
         // tile choice: empirically swept (tests/gpu sweep); override via
         // the `tile` arg (0 = auto heuristic)
         t = (int)tile;
+        if (t == 0) {
+            // experiment hook: PIPEEDGE_GEMM_T=<n> forces the auto choice
+            // (bench A/Bs without a rebuild); callers passing tile!=0 win.
+            static const int env_t = [] {
+                const char* s = std::getenv("PIPEEDGE_GEMM_T");
+                return s ? std::atoi(s) : 0;
+            }();
+            t = env_t;
+        }
         if (t == 0) {
             // from the on-HW sweeps (tools/gemm_sweep.py, hotpath_sweep.py,
             // glds3_screen.py): XCD-swizzled 64x64 2-buffer default; the
