@@ -533,7 +533,7 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
     }
     torch::Tensor stats;
     if (want_stats) {
-        const int bn_tile = t == 3 ? 128 : 64;
+        const int bn_tile = (t == 3 || t == 9) ? 128 : 64;
         const int parts = ((N + bn_tile - 1) / bn_tile) * 2;
         stats = torch::empty({M, parts, 2},
                              a2d.options().dtype(torch::kFloat32));
@@ -560,20 +560,33 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
     } while (0)
     if (K % BK == 0) {
+#define PE_LAUNCH_GLDS3(MI_, NI_, GRID_)                                     \
+    do {                                                                     \
+        if (epm == 1)                                                        \
+            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 1>),  \
+                               GRID_, dim3(256), 0, pe_current_stream(),     \
+                               ap, bp, cp, M, N, K, ep);                     \
+        else if (epm == 2)                                                   \
+            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 2>),  \
+                               GRID_, dim3(256), 0, pe_current_stream(),     \
+                               ap, bp, cp, M, N, K, ep);                     \
+        else                                                                 \
+            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 0>),  \
+                               GRID_, dim3(256), 0, pe_current_stream(),     \
+                               ap, bp, cp, M, N, K, ep);                     \
+    } while (0)
         if (t == 7) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-            if (epm == 1)
-                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 1>),
-                                   grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, cp, M, N, K, ep);
-            else if (epm == 2)
-                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 2>),
-                                   grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, cp, M, N, K, ep);
-            else
-                hipLaunchKernelGGL((gemm_bf16_glds3_kernel<2, 2, true, 0>),
-                                   grid, dim3(256), 0, pe_current_stream(),
-                                   ap, bp, cp, M, N, K, ep);
+            PE_LAUNCH_GLDS3(2, 2, grid);
+        } else if (t == 8) {
+            // 128x64 3-buffer: double arithmetic intensity vs t7 (the 64x64
+            // tiles are HBM/L2-bound at ~32 FLOP/B); 72 KiB LDS -> 2/CU
+            dim3 grid((M + 127) / 128, (N + BN - 1) / BN);
+            PE_LAUNCH_GLDS3(4, 2, grid);
+        } else if (t == 9) {
+            // 128x128 3-buffer: 96 KiB LDS -> 1/CU
+            dim3 grid((M + 127) / 128, (N + 127) / 128);
+            PE_LAUNCH_GLDS3(4, 4, grid);
         } else if (t == 4 || t == 5 || t == 6) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             if (t == 4) PE_LAUNCH_GLDS(2, 2, 2, false, grid);
@@ -605,6 +618,7 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                                ep);
     }
 #undef PE_LAUNCH_GLDS
+#undef PE_LAUNCH_GLDS3
     HIP_CHECK_LAST();
     return {out, stats};
 }

@@ -50,6 +50,20 @@ def test_gemm_matches_fp32(m, n, k):
     _assert_close(out.cpu(), ref, atol=0.05 * k ** 0.5 * 0.05)
 
 
+@pytest.mark.parametrize("tile", [1, 2, 3, 4, 5, 6, 7, 8, 9])
+def test_gemm_explicit_tiles(tile):
+    """Every dispatchable tile variant (incl. the 3-buffer 128x64/128x128
+    glds3 instantiations) against the fp32 reference on a K%64==0 shape."""
+    torch.manual_seed(tile)
+    a = torch.randn(300, 128) * 0.5
+    b = torch.randn(192, 128) * 0.1
+    bias = torch.randn(192)
+    ref = torch_ref.linear(a, b, bias)
+    out = _hip().linear(a.bfloat16().cuda(), b.bfloat16().cuda(),
+                        bias.bfloat16().cuda(), tile=tile)[0]
+    _assert_close(out.cpu(), ref, atol=0.05 * 128 ** 0.5 * 0.05)
+
+
 def test_gemm_gelu_residual():
     torch.manual_seed(0)
     a = torch.randn(197, 768) * 0.5

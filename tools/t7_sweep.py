@@ -1,5 +1,6 @@
-"""t5 (2-buffer XCD 64x64) vs t7 (3-buffer raw-barrier) on the hot GEMM
-shapes, to refine the auto-tile heuristic (gemm.hip host dispatch)."""
+"""Tile-variant micro-sweep on the hot GEMM shapes: t5 (2-buffer XCD 64x64),
+t7 (3-buffer 64x64), t8 (3-buffer 128x64), t9 (3-buffer 128x128) - refines
+the auto-tile heuristic in the gemm.hip host dispatch."""
 import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, time
 from pipeedge_amd.ops import hip_loader
@@ -20,6 +21,7 @@ shapes = [
     ("bert-qkv", 1024, 2304, 768),
     ("bert-fc2", 1024, 768, 3072),
 ]
+TILES = (5, 7, 8, 9)
 
 def bench(fn, iters=60):
     for _ in range(12):
@@ -30,12 +32,25 @@ def bench(fn, iters=60):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters * 1e6
 
-print(f"{'shape':10s} {'M':>5} {'N':>5} {'K':>5} |    t5us    t7us | win   TF(best)")
+hdr = " ".join(f"t{t}us".rjust(8) for t in TILES)
+print(f"{'shape':10s} {'M':>5} {'N':>5} {'K':>5} | {hdr} | win   TF(best)")
 for name, m, n, k in shapes:
     a = (torch.randn(m, k) * 0.3).bfloat16().cuda()
     b = (torch.randn(n, k) * 0.1).bfloat16().cuda()
-    r5 = bench(lambda: hip.linear(a, b, tile=5))
-    r7 = bench(lambda: hip.linear(a, b, tile=7))
-    best = min(r5, r7)
-    win = "t7" if r7 < r5 else "t5"
-    print(f"{name:10s} {m:>5} {n:>5} {k:>5} | {r5:7.1f} {r7:7.1f} | {win}  {2*m*n*k/best/1e6:6.0f} TF")
+    # correctness vs fp32 torch before timing
+    ref = a.float().matmul(b.float().t())
+    res = {}
+    for t in TILES:
+        out = hip.linear(a, b, tile=t)
+        out = out[0] if isinstance(out, (tuple, list)) else out
+        err = (out.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1e-6)
+        assert err < 2e-2, f"{name} t{t} rel err {err}"
+        # dummy pass to absorb allocator growth before the timed loop
+        for _ in range(3):
+            hip.linear(a, b, tile=t)
+        torch.cuda.synchronize()
+        res[t] = bench(lambda t=t: hip.linear(a, b, tile=t))
+    best_t = min(res, key=res.get)
+    cols = " ".join(f"{res[t]:8.1f}" for t in TILES)
+    print(f"{name:10s} {m:>5} {n:>5} {k:>5} | {cols} | t{best_t}  "
+          f"{2*m*n*k/res[best_t]/1e6:6.0f} TF")
