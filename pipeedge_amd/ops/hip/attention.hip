@@ -23,7 +23,13 @@ namespace {
 constexpr int QCH = 64;   // q rows per workgroup
 constexpr int PAD = 8;
 
-template <int DHP, int KVT = 64>
+// EXACT=true specializes dh == DHP at compile time (the ViT/BERT dh=64 hot
+// case): the ragged-column guards and their exec-mask branches vanish, and
+// K/V global addresses are hoisted to per-thread pointers advanced by
+// KVT*row_stride per tile instead of recomputed 64-bit mads (the disassembly
+// showed ~200 VALU address ops per tile; fresh PMC: MfmaUtil 2.9%,
+// VALUBusy 21%, i.e. the kernel was VALU/latency-bound, not memory-bound).
+template <int DHP, int KVT = 64, bool EXACT = false>
 __global__ __launch_bounds__(256) void qkv_attention_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
         int B, int S, int H, int dh, float scale) {
@@ -74,8 +80,8 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
         const int c = (piece % (DHP / 8)) * 8;
         bf16x8 v = {};
         const int gs = q0 + r;
-        if (gs < S && c < dh) {
-            if (c + 7 < dh) {
+        if (gs < S && (EXACT || c < dh)) {
+            if (EXACT || c + 7 < dh) {
                 v = *reinterpret_cast<const bf16x8*>(
                     q_base + (long)gs * row_stride + c);
             } else {
@@ -94,42 +100,50 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     // loads issue BEFORE tile t's compute so HBM latency hides under the
     // MFMA phase; the LDS writes happen after the next barrier.
     constexpr int PIECES = KVT * DHP / 8 / 256;  // vec8 pieces per thread
+    constexpr int RPT = 256 / (DHP / 8);         // tile rows per 256 threads
     bf16x8 k_reg[PIECES], v_reg[PIECES];
+
+    // per-thread K/V addresses computed once, advanced by KVT rows per tile
+    // (load_tile is called for t = 0, 1, ... exactly once each, in order)
+    const int lr0 = tid / (DHP / 8);
+    const int lc0 = (tid % (DHP / 8)) * 8;
+    const __bf16* k_pc[PIECES];
+    const __bf16* v_pc[PIECES];
+    #pragma unroll
+    for (int i = 0; i < PIECES; ++i) {
+        k_pc[i] = k_base + (long)(lr0 + i * RPT) * row_stride + lc0;
+        v_pc[i] = v_base + (long)(lr0 + i * RPT) * row_stride + lc0;
+    }
 
     auto load_tile = [&](int t) {
         const int s0 = t * KVT;
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int piece = tid + i * 256;
-            const int r = piece / (DHP / 8);
-            const int c = (piece % (DHP / 8)) * 8;
-            const int gs = s0 + r;
+            const int gs = s0 + lr0 + i * RPT;
             k_reg[i] = bf16x8{};
             v_reg[i] = bf16x8{};
-            if (gs < S && c < dh) {
-                if (c + 7 < dh) {
-                    k_reg[i] = *reinterpret_cast<const bf16x8*>(
-                        k_base + (long)gs * row_stride + c);
-                    v_reg[i] = *reinterpret_cast<const bf16x8*>(
-                        v_base + (long)gs * row_stride + c);
+            if (gs < S && (EXACT || lc0 < dh)) {
+                if (EXACT || lc0 + 7 < dh) {
+                    k_reg[i] = *reinterpret_cast<const bf16x8*>(k_pc[i]);
+                    v_reg[i] = *reinterpret_cast<const bf16x8*>(v_pc[i]);
                 } else {
-                    for (int j = 0; j < 8 && c + j < dh; ++j) {
-                        k_reg[i][j] = k_base[(long)gs * row_stride + c + j];
-                        v_reg[i][j] = v_base[(long)gs * row_stride + c + j];
+                    for (int j = 0; j < 8 && lc0 + j < dh; ++j) {
+                        k_reg[i][j] = k_pc[i][j];
+                        v_reg[i][j] = v_pc[i][j];
                     }
                 }
             }
+            k_pc[i] += (long)KVT * row_stride;
+            v_pc[i] += (long)KVT * row_stride;
         }
     };
     auto write_tile = [&]() {
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int piece = tid + i * 256;
-            const int r = piece / (DHP / 8);
-            const int c = (piece % (DHP / 8)) * 8;
-            *reinterpret_cast<bf16x8*>(&Ks[r][c]) = k_reg[i];
+            const int r = lr0 + i * RPT;
+            *reinterpret_cast<bf16x8*>(&Ks[r][lc0]) = k_reg[i];
             #pragma unroll
-            for (int j = 0; j < 8; ++j) *vs_at(c + j, r) = v_reg[i][j];
+            for (int j = 0; j < 8; ++j) *vs_at(lc0 + j, r) = v_reg[i][j];
         }
     };
 
@@ -219,17 +233,20 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     }
 
     // ---- epilogue: normalize and store ctx rows ----
+    const int srow0 = q0 + wid * 16 + (lane >> 4) * 4;
+    __bf16* out0 = ctx + ((long)b * S + srow0) * D + h * dh + (lane & 15);
+    float rdenom[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r)
+        rdenom[r] = 1.0f / (l_run[r] > 0.0f ? l_run[r] : 1.0f);
     #pragma unroll
     for (int nfo = 0; nfo < DHP / 16; ++nfo) {
         const int d = nfo * 16 + (lane & 15);
-        if (d >= dh) continue;
+        if (!EXACT && d >= dh) continue;
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            const int srow = q0 + wid * 16 + (lane >> 4) * 4 + r;
-            if (srow >= S) continue;
-            const float denom = l_run[r] > 0.0f ? l_run[r] : 1.0f;
-            ctx[((long)b * S + srow) * D + h * dh + d] =
-                f2bf(o_acc[nfo][r] / denom);
+            if (srow0 + r >= S) continue;
+            out0[(long)r * D + nfo * 16] = f2bf(o_acc[nfo][r] * rdenom[r]);
         }
     }
 }
@@ -448,16 +465,20 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         hipLaunchKernelGGL((qkv_attention_smallS_kernel<64, 256>), grid,
                            dim3(256), 0, pe_current_stream(), in, out,
                            B, S, H, dh, (float)scale);
-    } else if (dh <= 64) {
+    } else if (dh == 64) {
         if (kvt >= 128 && S >= 128) {
-            hipLaunchKernelGGL((qkv_attention_kernel<64, 128>), grid,
+            hipLaunchKernelGGL((qkv_attention_kernel<64, 128, true>), grid,
                                dim3(256), 0, pe_current_stream(), in, out,
                                B, S, H, dh, (float)scale);
         } else {
-            hipLaunchKernelGGL((qkv_attention_kernel<64, 64>), grid,
+            hipLaunchKernelGGL((qkv_attention_kernel<64, 64, true>), grid,
                                dim3(256), 0, pe_current_stream(), in, out,
                                B, S, H, dh, (float)scale);
         }
+    } else if (dh < 64) {
+        hipLaunchKernelGGL((qkv_attention_kernel<64, 64>), grid,
+                           dim3(256), 0, pe_current_stream(), in, out,
+                           B, S, H, dh, (float)scale);
     } else if (dh <= 96) {
         hipLaunchKernelGGL((qkv_attention_kernel<96>), grid, dim3(256), 0,
                            pe_current_stream(), in, out, B, S, H, dh,
