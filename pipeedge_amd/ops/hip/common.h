@@ -73,6 +73,23 @@ struct SumOp {
     __device__ float operator()(float a, float b) const { return a + b; }
 };
 
+// 32-lane (half-wave) reduction: DPP within the 16-lane rows, one
+// cross-row exchange. xor-16 stays inside each 32-lane half.
+template <typename Op>
+__device__ __forceinline__ float half32_reduce(float v, Op op) {
+    v = group16_reduce(v, op);
+    return op(v, __shfl_xor(v, 16));
+}
+
+// full 64-lane reduction: 4 DPP steps + 2 bpermute exchanges (vs 6
+// bpermute exchanges for a pure __shfl_xor ladder)
+template <typename Op>
+__device__ __forceinline__ float wave64_reduce(float v, Op op) {
+    v = group16_reduce(v, op);
+    v = op(v, __shfl_xor(v, 16));
+    return op(v, __shfl_xor(v, 32));
+}
+
 #define HIP_CHECK_LAST()                                                     \
     do {                                                                     \
         hipError_t e_ = hipGetLastError();                                   \

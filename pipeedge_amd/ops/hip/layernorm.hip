@@ -50,11 +50,8 @@ __global__ __launch_bounds__(256) void layernorm_reg_kernel(
             }
         }
     }
-    #pragma unroll
-    for (int m = 32; m >= 1; m >>= 1) {
-        sum += __shfl_xor(sum, m);
-        sumsq += __shfl_xor(sumsq, m);
-    }
+    sum = wave64_reduce(sum, SumOp());
+    sumsq = wave64_reduce(sumsq, SumOp());
     const float mean = sum / D;
     const float rstd = rsqrtf(sumsq / D - mean * mean + eps);
 
@@ -138,6 +135,64 @@ __global__ __launch_bounds__(256) void layernorm_split_kernel(
     }
 }
 
+// 8-rows-per-block variant: one 32-lane half-wave per row. Halves the
+// number of dispatch rounds at transformer row counts (1576 rows: 394
+// blocks -> 197, i.e. two HBM-latency rounds -> one); D <= NV * 256.
+template <bool WITH_RESIDUAL, int NV>
+__global__ __launch_bounds__(256) void layernorm_reg8_kernel(
+        const __bf16* __restrict__ x, const __bf16* __restrict__ residual,
+        const __bf16* __restrict__ weight, const __bf16* __restrict__ bias,
+        __bf16* __restrict__ out, long rows, int D, float eps) {
+    const long row = blockIdx.x * 8 + (threadIdx.x >> 5);
+    if (row >= rows) return;
+    const int lane = threadIdx.x & 31;
+    const __bf16* xr = x + row * D;
+    const __bf16* rr = WITH_RESIDUAL ? residual + row * D : nullptr;
+
+    float vals[NV][8];
+    float sum = 0.0f, sumsq = 0.0f;
+    #pragma unroll
+    for (int j = 0; j < NV; ++j) {
+        const int i = (j * 32 + lane) * 8;
+        if (i < D) {
+            bf16x8 v = *reinterpret_cast<const bf16x8*>(xr + i);
+            if (WITH_RESIDUAL) {
+                bf16x8 r = *reinterpret_cast<const bf16x8*>(rr + i);
+                #pragma unroll
+                for (int k = 0; k < 8; ++k)
+                    vals[j][k] = bf2f(v[k]) + bf2f(r[k]);
+            } else {
+                #pragma unroll
+                for (int k = 0; k < 8; ++k) vals[j][k] = bf2f(v[k]);
+            }
+            #pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                sum += vals[j][k];
+                sumsq += vals[j][k] * vals[j][k];
+            }
+        }
+    }
+    sum = half32_reduce(sum, SumOp());
+    sumsq = half32_reduce(sumsq, SumOp());
+    const float mean = sum / D;
+    const float rstd = rsqrtf(sumsq / D - mean * mean + eps);
+
+    #pragma unroll
+    for (int j = 0; j < NV; ++j) {
+        const int i = (j * 32 + lane) * 8;
+        if (i < D) {
+            bf16x8 w = *reinterpret_cast<const bf16x8*>(weight + i);
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(bias + i);
+            bf16x8 o;
+            #pragma unroll
+            for (int k = 0; k < 8; ++k)
+                o[k] = f2bf((vals[j][k] - mean) * rstd * bf2f(w[k]) +
+                            bf2f(b[k]));
+            *reinterpret_cast<bf16x8*>(out + row * D + i) = o;
+        }
+    }
+}
+
 // two-pass fallback for very wide rows
 template <bool WITH_RESIDUAL>
 __global__ __launch_bounds__(256) void layernorm_wide_kernel(
@@ -195,6 +250,31 @@ void launch_ln(const __bf16* x, const __bf16* res, const __bf16* w,
         return e != nullptr && std::atoi(e) != 0;  // default OFF (HW sweep:
         // the 1-wave register path wins at transformer row counts)
     }();
+    static const int rows_per_block = []() {
+        const char* e = std::getenv("PIPEEDGE_LN_R");
+        return e ? std::atoi(e) : 4;
+    }();
+    if (rows_per_block == 8 && D <= 1024) {
+        const dim3 grid8((rows + 7) / 8);
+        if (D <= 256)
+            hipLaunchKernelGGL((layernorm_reg8_kernel<WITH_RESIDUAL, 1>),
+                               grid8, block, 0, stream, x, res, w, b, out,
+                               rows, D, eps);
+        else if (D <= 512)
+            hipLaunchKernelGGL((layernorm_reg8_kernel<WITH_RESIDUAL, 2>),
+                               grid8, block, 0, stream, x, res, w, b, out,
+                               rows, D, eps);
+        else if (D <= 768)
+            hipLaunchKernelGGL((layernorm_reg8_kernel<WITH_RESIDUAL, 3>),
+                               grid8, block, 0, stream, x, res, w, b, out,
+                               rows, D, eps);
+        else
+            hipLaunchKernelGGL((layernorm_reg8_kernel<WITH_RESIDUAL, 4>),
+                               grid8, block, 0, stream, x, res, w, b, out,
+                               rows, D, eps);
+        HIP_CHECK_LAST();
+        return;
+    }
     if (split && D % 16 == 0 && D / 2 <= 1024) {
         // 2 waves per row, 2 rows per block
         hipLaunchKernelGGL((layernorm_split_kernel<WITH_RESIDUAL>),
