@@ -36,7 +36,15 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     __shared__ __bf16 Qs[QCH][DHP + PAD];
     __shared__ __bf16 Ks[KVT][DHP + PAD];
     __shared__ __bf16 Vs[DHP][KVT + PAD];      // transposed, XOR-swizzled
-    __shared__ __bf16 Ps[4][16][KVT + PAD];    // per-wave P tile
+    // The per-wave P tile OVERLAYS Ks: scores are done reading Ks before P
+    // is written (enforced by the barrier ahead of the P writes), and the
+    // loop-top barrier orders each tile's PV reads before the next K write.
+    // Cuts LDS 62->45 KiB at KVT=128 (2->3 blocks/CU) and 37->28 KiB at
+    // KVT=64 (4->5): the kernel is latency-bound (MfmaUtil 2.9%), so
+    // co-resident waves are the lever.
+    static_assert(4 * 16 * (KVT + PAD) <= KVT * (DHP + PAD),
+                  "P tile must fit inside Ks");
+    __bf16* Ps = &Ks[0][0];
 
     // Vs byte-offset swizzle: the transpose-staging writes put 8 lanes with
     // d-rows 8 apart on ONE bank (write modulus is 32 dwords; the aligned
@@ -58,7 +66,7 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
         int byte = ((w * 16 + row) * (KVT + PAD) + col) * 2;
         byte ^= ((row >> 3) & 1) << 5;
         return reinterpret_cast<__bf16*>(
-            reinterpret_cast<char*>(&Ps[0][0][0]) + byte);
+            reinterpret_cast<char*>(Ps) + byte);
     };
 
     const int bh = blockIdx.x;
@@ -207,7 +215,9 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
             for (int nfo = 0; nfo < DHP / 16; ++nfo)
                 o_acc[nfo][r] *= alpha;
         }
-        // P -> wave-private LDS (C layout -> A layout)
+        // P -> wave-private LDS (C layout -> A layout). P overlays Ks:
+        // every wave must be done reading Ks (scores) before any P write.
+        __syncthreads();
         #pragma unroll
         for (int nf = 0; nf < KVT / 16; ++nf)
             #pragma unroll
@@ -451,7 +461,11 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     auto* out = reinterpret_cast<__bf16*>(ctx.data_ptr());
     static const int kvt = []() {
         const char* e = std::getenv("PIPEEDGE_ATTN_KVT");
-        return e ? std::atoi(e) : 128;  // on-HW sweep: 128 wins for S >= 128
+        // KVT=128 is ~1us faster SOLO, but KVT=64 (28 KiB LDS after the
+        // P/Ks overlay -> 5 blocks/CU) wins the 2-stream e2e bench twice
+        // (10,134/10,024 vs 9,861/9,953 img/s): the kernel is latency-bound
+        // and co-residency with the other stream's GEMMs is the lever.
+        return e ? std::atoi(e) : 64;
     }();
     static const bool small_s = []() {
         // default OFF: on-HW A/B measured the whole-KV single-barrier
