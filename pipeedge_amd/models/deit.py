@@ -39,6 +39,7 @@ class DeiTEmbeddings(nn.Module):
         self.cls_token = _param((1, 1, d))
         self.dist_token = _param((1, 1, d))
         self.pos_embed = _param((1, config.seq_len, d))  # seq = patches + 2
+        self._prefix = None  # cached [2, d] cls+dist rows (built on first use)
 
     def init_random(self, prefix: str):
         for name, p in self.named_parameters():
@@ -48,10 +49,13 @@ class DeiTEmbeddings(nn.Module):
     def forward(self, pixels: Tensor) -> Tensor:
         patches = ops.patchify(pixels, self.config.patch_size)
         x = ops.linear(patches, self.patch_weight, self.patch_bias)
-        cls = self.cls_token.expand(x.shape[0], -1, -1).to(x.dtype)
-        dist = self.dist_token.expand(x.shape[0], -1, -1).to(x.dtype)
-        x = torch.cat((cls, dist, x), dim=1)
-        return x + self.pos_embed.to(x.dtype)
+        prefix = self._prefix
+        if prefix is None or prefix.dtype != x.dtype \
+                or prefix.device != x.device:
+            prefix = torch.cat((self.cls_token.reshape(1, -1),
+                                self.dist_token.reshape(1, -1))).to(x.dtype)
+            self._prefix = prefix
+        return ops.embed_finish(x, prefix, self.pos_embed[0].to(x.dtype))
 
 
 class DeiTLayerShard(ViTLayerShard):

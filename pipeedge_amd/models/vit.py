@@ -92,9 +92,8 @@ class ViTEmbeddings(nn.Module):
     def forward(self, pixels: Tensor) -> Tensor:
         patches = ops.patchify(pixels, self.config.patch_size)
         x = ops.linear(patches, self.patch_weight, self.patch_bias)
-        cls = self.cls_token.expand(x.shape[0], -1, -1).to(x.dtype)
-        x = torch.cat((cls, x), dim=1)
-        return x + self.pos_embed.to(x.dtype)
+        return ops.embed_finish(x, self.cls_token.reshape(1, -1).to(x.dtype),
+                                self.pos_embed[0].to(x.dtype))
 
 
 class ViTLayerShard(ModuleShard):

@@ -168,3 +168,11 @@ def patchify(pixels: Tensor, patch_size: int) -> Tensor:
     if _use_hip(pixels):
         return _hip.patchify(pixels, patch_size)
     return torch_ref.patchify(pixels, patch_size)
+
+
+def embed_finish(x: Tensor, prefix: Tensor, pos: Tensor) -> Tensor:
+    """Prepend prefix token rows (cls / cls+dist) and add position
+    embeddings: out[b, s] = (prefix[s] if s < T else x[b, s-T]) + pos[s]."""
+    if _use_hip(x):
+        return _hip.embed_finish(x, prefix, pos)
+    return torch_ref.embed_finish(x, prefix, pos)

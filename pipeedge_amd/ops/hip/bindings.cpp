@@ -22,6 +22,8 @@ torch::Tensor add_layernorm_bf16(torch::Tensor x, torch::Tensor residual,
 torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
                                  double scale);
 torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
+torch::Tensor embed_finish_bf16(torch::Tensor x, torch::Tensor prefix,
+                                torch::Tensor pos);
 std::vector<torch::Tensor> quant_encode_bf16(torch::Tensor x, int64_t bit,
                                              int64_t pad_words);
 torch::Tensor quant_decode_bf16(torch::Tensor comm, torch::Tensor scale,
@@ -79,6 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("attn_mask") = py::none());
     m.def("patchify", &patchify_bf16, py::arg("pixels"),
           py::arg("patch_size"));
+    m.def("embed_finish", &embed_finish_bf16, py::arg("x"), py::arg("prefix"),
+          py::arg("pos"));
     m.def("quant_encode", &quant_encode_bf16, py::arg("x"), py::arg("bit"),
           py::arg("pad_words") = 0);
     m.def("quant_decode", &quant_decode_bf16, py::arg("comm"),

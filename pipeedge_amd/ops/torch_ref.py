@@ -82,6 +82,13 @@ def qkv_attention(qkv: Tensor, num_heads: int, scale: Optional[float] = None,
     return ctx.permute(0, 2, 1, 3).reshape(b, s, d)
 
 
+def embed_finish(x: Tensor, prefix: Tensor, pos: Tensor) -> Tensor:
+    """out[b, s] = (prefix[s] if s < T else x[b, s-T]) + pos[s]."""
+    b = x.shape[0]
+    pre = prefix.unsqueeze(0).expand(b, -1, -1).to(x.dtype)
+    return torch.cat((pre, x), dim=1) + pos.to(x.dtype)
+
+
 def patchify(pixels: Tensor, patch_size: int) -> Tensor:
     """[B, C, H, W] -> [B, num_patches, C*P*P] for non-overlapping P x P patches.
 

@@ -270,3 +270,16 @@ def test_quant_hip_kernels_match_torch_path(bit):
     # cross-decode: HIP wire decoded by the torch path matches HIP decode
     cpu_dec = quant.tensor_decode_outerdim([t.cpu() for t in enc])
     assert (cpu_dec - dec.float().cpu()).abs().max().item() < 0.05
+
+
+@pytest.mark.parametrize("t", [1, 2])
+def test_embed_finish(t):
+    """Fused cls/dist-prepend + pos-add vs the torch reference."""
+    torch.manual_seed(t)
+    x = torch.randn(4, 196, 768) * 0.5
+    prefix = torch.randn(t, 768) * 0.5
+    pos = torch.randn(196 + t, 768) * 0.02
+    ref = torch_ref.embed_finish(x, prefix, pos)
+    out = _hip().embed_finish(x.bfloat16().cuda(), prefix.bfloat16().cuda(),
+                              pos.bfloat16().cuda())
+    _assert_close(out.cpu(), ref, atol=0.03)
