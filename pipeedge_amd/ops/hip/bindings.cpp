@@ -24,6 +24,9 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
 torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
 torch::Tensor embed_finish_bf16(torch::Tensor x, torch::Tensor prefix,
                                 torch::Tensor pos);
+torch::Tensor debug_mfma32_raw(torch::Tensor a, torch::Tensor b);
+torch::Tensor debug_tr16_raw();
+torch::Tensor debug_permlane32_raw();
 std::vector<torch::Tensor> quant_encode_bf16(torch::Tensor x, int64_t bit,
                                              int64_t pad_words);
 torch::Tensor quant_decode_bf16(torch::Tensor comm, torch::Tensor scale,
@@ -89,4 +92,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("scale"), py::arg("shift"), py::arg("bit"),
           py::arg("image_shape"));
     m.def("_debug_mfma_tile", &debug_mfma_tile);
+    m.def("_debug_mfma32_raw", &debug_mfma32_raw);
+    m.def("_debug_tr16_raw", &debug_tr16_raw);
+    m.def("_debug_permlane32_raw", &debug_permlane32_raw);
 }

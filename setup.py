@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "attention.hip"),
         os.path.join(HIP_DIR, "patchify.hip"),
         os.path.join(HIP_DIR, "quantpack.hip"),
+        os.path.join(HIP_DIR, "debug_probe.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
