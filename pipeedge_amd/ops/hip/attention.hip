@@ -108,34 +108,38 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     // loads issue BEFORE tile t's compute so HBM latency hides under the
     // MFMA phase; the LDS writes happen after the next barrier.
     constexpr int PIECES = KVT * DHP / 8 / 256;  // vec8 pieces per thread
-    constexpr int RPT = 256 / (DHP / 8);         // tile rows per 256 threads
     bf16x8 k_reg[PIECES], v_reg[PIECES];
 
     // per-thread K/V addresses computed once, advanced by KVT rows per tile
-    // (load_tile is called for t = 0, 1, ... exactly once each, in order)
-    const int lr0 = tid / (DHP / 8);
-    const int lc0 = (tid % (DHP / 8)) * 8;
+    // (load_tile is called for t = 0, 1, ... exactly once each, in order).
+    // Row/col come from the bijective piece index tid + i*256 - NOT from
+    // tid/(DHP/8) + i*(256/(DHP/8)), which truncates when DHP/8 does not
+    // divide 256 (DHP=96 left K/V row 63 cols 32.. unstaged -> stale LDS).
+    int lr[PIECES], lc[PIECES];
     const __bf16* k_pc[PIECES];
     const __bf16* v_pc[PIECES];
     #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
-        k_pc[i] = k_base + (long)(lr0 + i * RPT) * row_stride + lc0;
-        v_pc[i] = v_base + (long)(lr0 + i * RPT) * row_stride + lc0;
+        const int piece = tid + i * 256;
+        lr[i] = piece / (DHP / 8);
+        lc[i] = (piece % (DHP / 8)) * 8;
+        k_pc[i] = k_base + (long)lr[i] * row_stride + lc[i];
+        v_pc[i] = v_base + (long)lr[i] * row_stride + lc[i];
     }
 
     auto load_tile = [&](int t) {
         const int s0 = t * KVT;
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int gs = s0 + lr0 + i * RPT;
+            const int gs = s0 + lr[i];
             k_reg[i] = bf16x8{};
             v_reg[i] = bf16x8{};
-            if (gs < S && (EXACT || lc0 < dh)) {
-                if (EXACT || lc0 + 7 < dh) {
+            if (gs < S && (EXACT || lc[i] < dh)) {
+                if (EXACT || lc[i] + 7 < dh) {
                     k_reg[i] = *reinterpret_cast<const bf16x8*>(k_pc[i]);
                     v_reg[i] = *reinterpret_cast<const bf16x8*>(v_pc[i]);
                 } else {
-                    for (int j = 0; j < 8 && lc0 + j < dh; ++j) {
+                    for (int j = 0; j < 8 && lc[i] + j < dh; ++j) {
                         k_reg[i][j] = k_pc[i][j];
                         v_reg[i][j] = v_pc[i][j];
                     }
@@ -148,10 +152,9 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
     auto write_tile = [&]() {
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int r = lr0 + i * RPT;
-            *reinterpret_cast<bf16x8*>(&Ks[r][lc0]) = k_reg[i];
+            *reinterpret_cast<bf16x8*>(&Ks[lr[i]][lc[i]]) = k_reg[i];
             #pragma unroll
-            for (int j = 0; j < 8; ++j) *vs_at(lc0 + j, r) = v_reg[i][j];
+            for (int j = 0; j < 8; ++j) *vs_at(lc[i] + j, lr[i]) = v_reg[i][j];
         }
     };
 
@@ -258,6 +261,273 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
             if (srow0 + r >= S) continue;
             out0[(long)r * D + nfo * 16] = f2bf(o_acc[nfo][r] * rdenom[r]);
         }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// v2: swapped-operand flash attention (guide section B "fused attention
+// prefill", idioms T12/tr-read; dh = 64 only). Differences from v1:
+//   - QK^T computed as mfma(K, Q) with v_mfma_f32_32x32x16_bf16, so each
+//     lane holds 32 of the tile's 64 scores for ONE q row (q = lane&31;
+//     partner lane q^32 holds the rest): the whole online softmax runs in
+//     registers - no cross-lane DPP reduces, no P round-trip through LDS.
+//   - P converts to the PV B-fragment via v_cvt_pk_bf16_f32 +
+//     v_permlane32_swap (one swap fills two fragment dwords).
+//   - PV is computed as mfma(V^T, P) so the O accumulator keeps q = lane&31
+//     (the rescale factor is a per-lane scalar); the V^T A-fragment comes
+//     from ds_read_b64_tr_b16 over a [s/4][d/16][4][16]-blocked natural-V
+//     image - the transpose-write scalar storm of v1 is gone.
+// All lane mappings pinned by tools/probe_layouts.py on hardware
+// (profiles/layout_probes.txt). LDS: 16 KiB (K tile + V image).
+// ---------------------------------------------------------------------------
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef unsigned uint2_ __attribute__((ext_vector_type(2)));
+typedef __bf16 v4bf __attribute__((ext_vector_type(4)));
+typedef unsigned uint4_ __attribute__((ext_vector_type(4)));
+
+// exchange a value with the partner lane (l ^ 32) and combine. The copy
+// through v_mov is REQUIRED: v_permlane32_swap_b32 is a read-modify-write
+// exchange of TWO registers, so passing the same SSA value for both
+// operands lets the compiler use one register and the self-exchange
+// drops half the lanes (observed: running-max missed a hi-half spike ->
+// exp overflow -> inf outputs).
+// NOTE: do NOT use __builtin_amdgcn_permlane32_swap(v, v) here: with
+// value-identical operands the compiler folds the two swap results into
+// one (observed codegen: `v_max_f32 v81, v81, v81` after the swap), so
+// the partner half never enters the combine - the running max missed
+// hi-half spikes (-> exp overflow -> inf) and l_run doubled the own-half
+// sum. permlane stays in the P-pack path, where its operands differ.
+template <typename Op>
+__device__ __forceinline__ float partner32_combine(float v, Op op) {
+    return op(v, __shfl_xor(v, 32));
+}
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+    unsigned r;
+    // s_nop covers the VALU-write -> v_permlane-read hazard (guide T21):
+    // the consumer of this value is always a permlane32_swap
+    asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 0"
+                 : "=v"(r)
+                 : "v"(lo), "v"(hi));
+    return r;
+}
+
+template <int KVT>
+__global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
+        const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
+        int B, int S, int H, float scale) {
+    constexpr int DH = 64;
+    constexpr int QB = 128;                  // q rows per block (4 x 32)
+    __shared__ __bf16 Ks[KVT][DH];           // row-major, XOR-swizzled cols
+    __shared__ __bf16 Vs[KVT * DH];          // [s/4][d/16][4][16] blocked
+
+    const int bh = blockIdx.x;
+    const int b = bh / H;
+    const int h = bh % H;
+    const int q0 = blockIdx.y * QB;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l31 = lane & 31;
+    const int hi = lane >> 5;                // k-half of the fragment
+    const int D = H * DH;
+    const long row_stride = 3L * D;
+    const __bf16* q_base = qkv + (long)b * S * row_stride + h * DH;
+    const __bf16* k_base = q_base + D;
+    const __bf16* v_base = q_base + 2L * D;
+
+    // ---- Q -> registers as 4 B-fragments, pre-scaled (scale = 2^-3 for
+    // dh 64: exact in bf16) ----
+    const int qrow = q0 + wid * 32 + l31;
+    bf16x8 bq[4];
+    {
+        const __bf16* qp = q_base + (long)min(qrow, S - 1) * row_stride;
+        #pragma unroll
+        for (int kc = 0; kc < 4; ++kc) {
+            bf16x8 v = {};
+            if (qrow < S)
+                v = *reinterpret_cast<const bf16x8*>(qp + kc * 16 + hi * 8);
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) v[j] = f2bf(bf2f(v[j]) * scale);
+            bq[kc] = v;
+        }
+    }
+
+    // ---- K/V staging: per-thread running pointers (as in v1) ----
+    constexpr int PIECES = KVT / 32;
+    const int r0 = tid >> 3;                 // row-in-tile (8 lanes/row)
+    const int c0 = (tid & 7) * 8;
+    const __bf16* k_pc[PIECES];
+    const __bf16* v_pc[PIECES];
+    #pragma unroll
+    for (int i = 0; i < PIECES; ++i) {
+        k_pc[i] = k_base + (long)(r0 + i * 32) * row_stride + c0;
+        v_pc[i] = v_base + (long)(r0 + i * 32) * row_stride + c0;
+    }
+    bf16x8 k_reg[PIECES], v_reg[PIECES];
+    auto load_tile = [&](int t) {
+        const int s0_ = t * KVT;
+        #pragma unroll
+        for (int i = 0; i < PIECES; ++i) {
+            const int gs = s0_ + r0 + i * 32;
+            k_reg[i] = bf16x8{};
+            v_reg[i] = bf16x8{};
+            if (gs < S) {
+                k_reg[i] = *reinterpret_cast<const bf16x8*>(k_pc[i]);
+                v_reg[i] = *reinterpret_cast<const bf16x8*>(v_pc[i]);
+            }
+            k_pc[i] += (long)KVT * row_stride;
+            v_pc[i] += (long)KVT * row_stride;
+        }
+    };
+    auto write_tile = [&]() {
+        #pragma unroll
+        for (int i = 0; i < PIECES; ++i) {
+            const int r = r0 + i * 32;
+            // K: swizzle the 16-B column chunk by row&7 (read side matches)
+            const int kchunk = (c0 / 8) ^ (r & 7);
+            *reinterpret_cast<bf16x8*>(&Ks[r][kchunk * 8]) = k_reg[i];
+            // V: blocked tile (s/4, d/16), row s%4, col d%16 - the thread's
+            // 8 d-elements are contiguous within one 16-wide tile row
+            const int vtile = (r / 4) * (DH / 16) + c0 / 16;
+            *reinterpret_cast<bf16x8*>(
+                &Vs[vtile * 64 + (r % 4) * 16 + (c0 % 16)]) = v_reg[i];
+        }
+    };
+
+    f32x16 o_acc[2] = {{}, {}};
+    float m_run = -1e30f, l_run = 0.0f;
+
+    const int n_tiles = (S + KVT - 1) / KVT;
+    load_tile(0);
+    for (int t = 0; t < n_tiles; ++t) {
+        const int s0 = t * KVT;
+        __syncthreads();             // prior tile's LDS reads complete
+        write_tile();
+        __syncthreads();             // staging visible
+        if (t + 1 < n_tiles) load_tile(t + 1);
+
+        // ---- scores: D[kpos][q] via mfma(K, Q); lane holds, for its q,
+        // kpos = mt*32 + hi*4 + (i&3) + 8*(i>>2) ----
+        f32x16 sfrag[KVT / 32];
+        #pragma unroll
+        for (int mt = 0; mt < KVT / 32; ++mt) sfrag[mt] = f32x16{};
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int kc = 0; kc < 4; ++kc) {
+            #pragma unroll
+            for (int mt = 0; mt < KVT / 32; ++mt) {
+                const int row = mt * 32 + l31;
+                const int chunk = (kc * 2 + hi) ^ (row & 7);
+                const bf16x8 ak = *reinterpret_cast<const bf16x8*>(
+                    &Ks[row][chunk * 8]);
+                sfrag[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    ak, bq[kc], sfrag[mt], 0, 0, 0);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+
+        if (s0 + KVT > S) {          // ragged tail: mask invalid kpos
+            #pragma unroll
+            for (int mt = 0; mt < KVT / 32; ++mt)
+                #pragma unroll
+                for (int i = 0; i < 16; ++i)
+                    if (s0 + mt * 32 + hi * 4 + (i & 3) + 8 * (i >> 2) >= S)
+                        sfrag[mt][i] = -1e30f;
+        }
+
+        // ---- online softmax, fully in registers (q = lane&31; partner
+        // lane q^32 holds the other 32 kpos) ----
+        float tmax = -1e30f;
+        #pragma unroll
+        for (int mt = 0; mt < KVT / 32; ++mt)
+            #pragma unroll
+            for (int i = 0; i < 16; ++i) tmax = fmaxf(tmax, sfrag[mt][i]);
+        tmax = partner32_combine(tmax, MaxOp());
+        const float m_new = fmaxf(m_run, tmax);
+        const float alpha = __expf(m_run - m_new);
+        float rsum = 0.0f;
+        #pragma unroll
+        for (int mt = 0; mt < KVT / 32; ++mt)
+            #pragma unroll
+            for (int i = 0; i < 16; ++i) {
+                const float p = __expf(sfrag[mt][i] - m_new);
+                sfrag[mt][i] = p;    // P replaces scores in-place
+                rsum += p;
+            }
+        rsum = partner32_combine(rsum, SumOp());
+        l_run = l_run * alpha + rsum;
+        m_run = m_new;
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+            #pragma unroll
+            for (int i = 0; i < 16; ++i) o_acc[nt][i] *= alpha;
+
+        // ---- PV: O[d][q] += V^T . P per 16-kpos chunk ----
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int kc2 = 0; kc2 < KVT / 16; ++kc2) {
+            const int a_ = kc2 & 1, mt = kc2 >> 1;
+            // P B-fragment: own quartets packed to bf16, halves exchanged
+            // with the partner lane (one swap fills dwords 0 and 2)
+            const unsigned u0 = cvt_pk_bf16(sfrag[mt][8 * a_ + 0],
+                                            sfrag[mt][8 * a_ + 1]);
+            const unsigned u1 = cvt_pk_bf16(sfrag[mt][8 * a_ + 2],
+                                            sfrag[mt][8 * a_ + 3]);
+            const unsigned v0 = cvt_pk_bf16(sfrag[mt][8 * a_ + 4],
+                                            sfrag[mt][8 * a_ + 5]);
+            const unsigned v1 = cvt_pk_bf16(sfrag[mt][8 * a_ + 6],
+                                            sfrag[mt][8 * a_ + 7]);
+            const uint2_ w0 = __builtin_amdgcn_permlane32_swap(u0, v0,
+                                                               false, false);
+            const uint2_ w1 = __builtin_amdgcn_permlane32_swap(u1, v1,
+                                                               false, false);
+            const uint4_ pw = {w0.x, w1.x, w0.y, w1.y};
+            const bf16x8 pb = __builtin_bit_cast(bf16x8, pw);
+            const int g = lane >> 4;
+            const int sb1 = kc2 * 4 + (g >> 1) * 2;
+            #pragma unroll
+            for (int nt = 0; nt < 2; ++nt) {
+                const int db = nt * 2 + (g & 1);
+                auto* t1 = reinterpret_cast<
+                    __attribute__((address_space(3))) v4bf*>(
+                    reinterpret_cast<uintptr_t>(&Vs[0]) +
+                    (sb1 * 4 + db) * 128 + (lane & 15) * 8);
+                auto* t2 = reinterpret_cast<
+                    __attribute__((address_space(3))) v4bf*>(
+                    reinterpret_cast<uintptr_t>(&Vs[0]) +
+                    ((sb1 + 1) * 4 + db) * 128 + (lane & 15) * 8);
+                const v4bf r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t1);
+                const v4bf r2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t2);
+                bf16x8 av;
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    av[j] = r1[j];
+                    av[j + 4] = r2[j];
+                }
+                o_acc[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    av, pb, o_acc[nt], 0, 0, 0);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- epilogue: O[d][q] -> ctx rows, normalized ----
+    if (qrow < S) {
+        const float rdenom = 1.0f / (l_run > 0.0f ? l_run : 1.0f);
+        __bf16* out0 = ctx + ((long)b * S + qrow) * D + h * DH;
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+            #pragma unroll
+            for (int i4 = 0; i4 < 4; ++i4) {
+                const int d0 = nt * 32 + hi * 4 + 8 * i4;
+                v4bf o4;
+                #pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    o4[j] = f2bf(o_acc[nt][i4 * 4 + j] * rdenom);
+                *reinterpret_cast<v4bf*>(out0 + d0) = o4;
+            }
     }
 }
 
@@ -467,6 +737,20 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         // and co-residency with the other stream's GEMMs is the lever.
         return e ? std::atoi(e) : 64;
     }();
+    static const int v2 = []() {
+        // default ON for dh=64 (measured: 9.8us vs 10.4us solo, and e2e
+        // 10,300-10,500 vs ~10,000-10,200 img/s); set 0 to fall back to v1
+        const char* e = std::getenv("PIPEEDGE_ATTN_V2");
+        return e ? std::atoi(e) : 1;
+    }();
+    if (dh == 64 && v2) {
+        dim3 grid2(B * H, (S + 127) / 128);
+        hipLaunchKernelGGL((qkv_attention_v2_kernel<64>), grid2, dim3(256),
+                           0, pe_current_stream(), in, out, B, S, H,
+                           (float)scale);
+        HIP_CHECK_LAST();
+        return ctx;
+    }
     static const bool small_s = []() {
         // default OFF: on-HW A/B measured the whole-KV single-barrier
         // variant at 13.5us vs 11.5us for the tiled kernel at S=197 (its
