@@ -520,11 +520,15 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
         }
         if (t == 0) {
             // from the on-HW sweeps (tools/gemm_sweep.py, hotpath_sweep.py,
-            // glds3_screen.py): XCD-swizzled 64x64 2-buffer default; the
-            // 3-buffer raw-barrier pipeline (t7) for long-K / wide-N where
-            // per-block serial latency dominates (fc1 18.1->17.0us,
-            // fc2 22.9->22.5us)
-            if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
+            // glds3_screen.py, bigshape_sweep.py): XCD-swizzled 64x64
+            // 2-buffer default; the 3-buffer pipeline (t7) for the
+            // mid-size long-K / wide-N shapes (ViT-B/L fc1/fc2 class);
+            // 128-row tiles once M >= 2048 (ViT-H class: t2 beats t7 by
+            // 15% on fc1, t5 beats t7 by 18% on fc2); 128x128 for the
+            // large near-square regime (4k^3: t3 is 2x t7)
+            if (M >= 4096 && N >= 4096 && K >= 2048) t = 3;
+            else if (M >= 2048) t = (N >= 3072) ? 2 : 5;
+            else if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
             else t = 5;
         }
         if (t == 4 && K % (2 * BK) != 0) t = 1;
