@@ -313,14 +313,20 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
     return r;
 }
 
-template <int KVT>
+// DHP: padded head dim (32-multiple; dh <= DHP, zero-padded). EXACT folds
+// the dh==DHP guards away (the ViT/BERT dh=64 hot case). KROW pads the K
+// LDS row to 128 elements for DHP > 64 so the row stride stays 256 B and
+// the 16-chunk XOR swizzle keeps the A-fragment reads conflict-free.
+template <int KVT, int DHP = 64, bool EXACT = true>
 __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
-        int B, int S, int H, float scale) {
-    constexpr int DH = 64;
+        int B, int S, int H, int dh, float scale) {
     constexpr int QB = 128;                  // q rows per block (4 x 32)
-    __shared__ __bf16 Ks[KVT][DH];           // row-major, XOR-swizzled cols
-    __shared__ __bf16 Vs[KVT * DH];          // [s/4][d/16][4][16] blocked
+    constexpr int KROW = (DHP <= 64) ? 64 : 128;   // padded K row (elems)
+    constexpr int SWZ = (DHP <= 64) ? 7 : 15;      // 16-B chunk XOR mask
+    constexpr int NT = DHP / 32;             // 32-wide d tiles
+    __shared__ __bf16 Ks[KVT][KROW];         // row-major, XOR-swizzled cols
+    __shared__ __bf16 Vs[KVT * DHP];         // [s/4][d/16][4][16] blocked
 
     const int bh = blockIdx.x;
     const int b = bh / H;
@@ -331,49 +337,50 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     const int wid = tid >> 6;
     const int l31 = lane & 31;
     const int hi = lane >> 5;                // k-half of the fragment
-    const int D = H * DH;
+    const int D = H * dh;
     const long row_stride = 3L * D;
-    const __bf16* q_base = qkv + (long)b * S * row_stride + h * DH;
+    const __bf16* q_base = qkv + (long)b * S * row_stride + h * dh;
     const __bf16* k_base = q_base + D;
     const __bf16* v_base = q_base + 2L * D;
 
-    // ---- Q -> registers as 4 B-fragments, pre-scaled (scale = 2^-3 for
-    // dh 64: exact in bf16) ----
+    // ---- Q -> registers as DHP/16 B-fragments (scores scale post-MFMA) ----
     const int qrow = q0 + wid * 32 + l31;
-    bf16x8 bq[4];
+    bf16x8 bq[DHP / 16];
     {
         const __bf16* qp = q_base + (long)min(qrow, S - 1) * row_stride;
         #pragma unroll
-        for (int kc = 0; kc < 4; ++kc) {
+        for (int kc = 0; kc < DHP / 16; ++kc) {
             bf16x8 v = {};
-            if (qrow < S)
-                v = *reinterpret_cast<const bf16x8*>(qp + kc * 16 + hi * 8);
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) v[j] = f2bf(bf2f(v[j]) * scale);
+            const int c = kc * 16 + hi * 8;
+            if (qrow < S && (EXACT || c + 7 < dh))
+                v = *reinterpret_cast<const bf16x8*>(qp + c);
             bq[kc] = v;
         }
     }
 
-    // ---- K/V staging: per-thread running pointers (as in v1) ----
-    constexpr int PIECES = KVT / 32;
-    const int r0 = tid >> 3;                 // row-in-tile (8 lanes/row)
-    const int c0 = (tid & 7) * 8;
+    // ---- K/V staging: per-thread running pointers; row/col from the
+    // bijective piece index (DHP/8 need not divide 256) ----
+    constexpr int PIECES = KVT * DHP / 8 / 256;
+    int lr[PIECES], lc[PIECES];
     const __bf16* k_pc[PIECES];
     const __bf16* v_pc[PIECES];
     #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
-        k_pc[i] = k_base + (long)(r0 + i * 32) * row_stride + c0;
-        v_pc[i] = v_base + (long)(r0 + i * 32) * row_stride + c0;
+        const int piece = tid + i * 256;
+        lr[i] = piece / (DHP / 8);
+        lc[i] = (piece % (DHP / 8)) * 8;
+        k_pc[i] = k_base + (long)lr[i] * row_stride + lc[i];
+        v_pc[i] = v_base + (long)lr[i] * row_stride + lc[i];
     }
     bf16x8 k_reg[PIECES], v_reg[PIECES];
     auto load_tile = [&](int t) {
         const int s0_ = t * KVT;
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int gs = s0_ + r0 + i * 32;
+            const int gs = s0_ + lr[i];
             k_reg[i] = bf16x8{};
             v_reg[i] = bf16x8{};
-            if (gs < S) {
+            if (gs < S && (EXACT || lc[i] + 7 < dh)) {
                 k_reg[i] = *reinterpret_cast<const bf16x8*>(k_pc[i]);
                 v_reg[i] = *reinterpret_cast<const bf16x8*>(v_pc[i]);
             }
@@ -384,19 +391,19 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     auto write_tile = [&]() {
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            const int r = r0 + i * 32;
-            // K: swizzle the 16-B column chunk by row&7 (read side matches)
-            const int kchunk = (c0 / 8) ^ (r & 7);
+            const int r = lr[i];
+            // K: swizzle the 16-B column chunk by row&SWZ (reads match)
+            const int kchunk = (lc[i] / 8) ^ (r & SWZ);
             *reinterpret_cast<bf16x8*>(&Ks[r][kchunk * 8]) = k_reg[i];
             // V: blocked tile (s/4, d/16), row s%4, col d%16 - the thread's
             // 8 d-elements are contiguous within one 16-wide tile row
-            const int vtile = (r / 4) * (DH / 16) + c0 / 16;
+            const int vtile = (r / 4) * (DHP / 16) + lc[i] / 16;
             *reinterpret_cast<bf16x8*>(
-                &Vs[vtile * 64 + (r % 4) * 16 + (c0 % 16)]) = v_reg[i];
+                &Vs[vtile * 64 + (r % 4) * 16 + (lc[i] % 16)]) = v_reg[i];
         }
     };
 
-    f32x16 o_acc[2] = {{}, {}};
+    f32x16 o_acc[NT] = {};
     float m_run = -1e30f, l_run = 0.0f;
 
     const int n_tiles = (S + KVT - 1) / KVT;
@@ -415,11 +422,11 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         for (int mt = 0; mt < KVT / 32; ++mt) sfrag[mt] = f32x16{};
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int kc = 0; kc < 4; ++kc) {
+        for (int kc = 0; kc < DHP / 16; ++kc) {
             #pragma unroll
             for (int mt = 0; mt < KVT / 32; ++mt) {
                 const int row = mt * 32 + l31;
-                const int chunk = (kc * 2 + hi) ^ (row & 7);
+                const int chunk = (kc * 2 + hi) ^ (row & SWZ);
                 const bf16x8 ak = *reinterpret_cast<const bf16x8*>(
                     &Ks[row][chunk * 8]);
                 sfrag[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -428,6 +435,10 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         }
         __builtin_amdgcn_s_setprio(0);
 
+        #pragma unroll
+        for (int mt = 0; mt < KVT / 32; ++mt)
+            #pragma unroll
+            for (int i = 0; i < 16; ++i) sfrag[mt][i] *= scale;
         if (s0 + KVT > S) {          // ragged tail: mask invalid kpos
             #pragma unroll
             for (int mt = 0; mt < KVT / 32; ++mt)
@@ -460,7 +471,7 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         l_run = l_run * alpha + rsum;
         m_run = m_new;
         #pragma unroll
-        for (int nt = 0; nt < 2; ++nt)
+        for (int nt = 0; nt < NT; ++nt)
             #pragma unroll
             for (int i = 0; i < 16; ++i) o_acc[nt][i] *= alpha;
 
@@ -488,16 +499,16 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
             const int g = lane >> 4;
             const int sb1 = kc2 * 4 + (g >> 1) * 2;
             #pragma unroll
-            for (int nt = 0; nt < 2; ++nt) {
+            for (int nt = 0; nt < NT; ++nt) {
                 const int db = nt * 2 + (g & 1);
                 auto* t1 = reinterpret_cast<
                     __attribute__((address_space(3))) v4bf*>(
                     reinterpret_cast<uintptr_t>(&Vs[0]) +
-                    (sb1 * 4 + db) * 128 + (lane & 15) * 8);
+                    (sb1 * (DHP / 16) + db) * 128 + (lane & 15) * 8);
                 auto* t2 = reinterpret_cast<
                     __attribute__((address_space(3))) v4bf*>(
                     reinterpret_cast<uintptr_t>(&Vs[0]) +
-                    ((sb1 + 1) * 4 + db) * 128 + (lane & 15) * 8);
+                    ((sb1 + 1) * (DHP / 16) + db) * 128 + (lane & 15) * 8);
                 const v4bf r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t1);
                 const v4bf r2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t2);
                 bf16x8 av;
@@ -516,12 +527,13 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     // ---- epilogue: O[d][q] -> ctx rows, normalized ----
     if (qrow < S) {
         const float rdenom = 1.0f / (l_run > 0.0f ? l_run : 1.0f);
-        __bf16* out0 = ctx + ((long)b * S + qrow) * D + h * DH;
+        __bf16* out0 = ctx + ((long)b * S + qrow) * D + h * dh;
         #pragma unroll
-        for (int nt = 0; nt < 2; ++nt)
+        for (int nt = 0; nt < NT; ++nt)
             #pragma unroll
             for (int i4 = 0; i4 < 4; ++i4) {
                 const int d0 = nt * 32 + hi * 4 + 8 * i4;
+                if (!EXACT && d0 >= dh) continue;
                 v4bf o4;
                 #pragma unroll
                 for (int j = 0; j < 4; ++j)
@@ -745,9 +757,23 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     }();
     if (dh == 64 && v2) {
         dim3 grid2(B * H, (S + 127) / 128);
-        hipLaunchKernelGGL((qkv_attention_v2_kernel<64>), grid2, dim3(256),
-                           0, pe_current_stream(), in, out, B, S, H,
-                           (float)scale);
+        hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 64, true>), grid2,
+                           dim3(256), 0, pe_current_stream(), in, out, B, S,
+                           H, dh, (float)scale);
+        HIP_CHECK_LAST();
+        return ctx;
+    }
+    if (dh > 64 && dh <= 96 && dh % 8 == 0 && v2) {
+        // ViT-Huge class (dh=80): zero-padded to DHP=96
+        dim3 grid2(B * H, (S + 127) / 128);
+        if (dh == 96)
+            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 96, true>),
+                               grid2, dim3(256), 0, pe_current_stream(), in,
+                               out, B, S, H, dh, (float)scale);
+        else
+            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 96, false>),
+                               grid2, dim3(256), 0, pe_current_stream(), in,
+                               out, B, S, H, dh, (float)scale);
         HIP_CHECK_LAST();
         return ctx;
     }

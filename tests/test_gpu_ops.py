@@ -283,3 +283,23 @@ def test_embed_finish(t):
     out = _hip().embed_finish(x.bfloat16().cuda(), prefix.bfloat16().cuda(),
                               pos.bfloat16().cuda())
     _assert_close(out.cpu(), ref, atol=0.03)
+
+
+def test_vit_huge_forward_hip_vs_cpu_fp32():
+    """ViT-Huge exercises the dh=80 (DHP=96 zero-padded) attention v2 path,
+    patch-14 patchify, and the M>=2048 GEMM tile class."""
+    from pipeedge_amd.models import ModuleShardConfig
+    from pipeedge_amd.models.config import get_config
+    from pipeedge_amd.models import vit
+    cfg = get_config("google/vit-huge-patch14-224-in21k")
+    sc = ModuleShardConfig(layer_start=1, layer_end=8, is_first=True,
+                           is_last=False)
+    model = vit.ViTModelShard(cfg, sc, None)
+    torch.manual_seed(0)
+    x = torch.randn(2, 3, 224, 224)
+    ref = model(x)  # CPU fp32
+    gpu_model = vit.ViTModelShard(cfg, sc, None).to("cuda", torch.bfloat16)
+    out = gpu_model(x.bfloat16().cuda())
+    cos = torch.nn.functional.cosine_similarity(
+        out.float().cpu().flatten(), ref.flatten(), dim=0)
+    assert cos.item() > 0.99, cos.item()
