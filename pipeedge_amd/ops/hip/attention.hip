@@ -757,9 +757,20 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     }();
     if (dh == 64 && v2) {
         dim3 grid2(B * H, (S + 127) / 128);
-        hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 64, true>), grid2,
-                           dim3(256), 0, pe_current_stream(), in, out, B, S,
-                           H, dh, (float)scale);
+        // KV tile length: 128 halves the barrier count and wins at ViT-class
+        // sequences (9.1 vs 9.7us solo; e2e 10,500), but its doubled score
+        // register state loses at long sequences (BERT s384: 5,290 vs
+        // 5,525) - auto-pick by S; PIPEEDGE_ATTN_V2=2 forces 128, =3
+        // forces 64
+        const bool kvt128 = v2 == 2 || (v2 == 1 && S > 64 && S <= 256);
+        if (kvt128)
+            hipLaunchKernelGGL((qkv_attention_v2_kernel<128, 64, true>),
+                               grid2, dim3(256), 0, pe_current_stream(), in,
+                               out, B, S, H, dh, (float)scale);
+        else
+            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 64, true>),
+                               grid2, dim3(256), 0, pe_current_stream(), in,
+                               out, B, S, H, dh, (float)scale);
         HIP_CHECK_LAST();
         return ctx;
     }
