@@ -266,7 +266,8 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
 
 // ---------------------------------------------------------------------------
 // v2: swapped-operand flash attention (guide section B "fused attention
-// prefill", idioms T12/tr-read; dh = 64 only). Differences from v1:
+// prefill", idioms T12/tr-read; head dims <= 96 via the DHP template,
+// zero-padded when dh < DHP). Differences from v1:
 //   - QK^T computed as mfma(K, Q) with v_mfma_f32_32x32x16_bf16, so each
 //     lane holds 32 of the tile's 64 scores for ONE q row (q = lane&31;
 //     partner lane q^32 holds the rest): the whole online softmax runs in
