@@ -274,20 +274,23 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// 3-buffer glds pipeline (K % 64 == 0): two K-tiles of loads stay in flight
-// ACROSS barriers (guide "Pipelining across barriers"): raw s_barrier +
-// counted `s_waitcnt vmcnt(N)` instead of __syncthreads(), whose fence
-// would drain the in-flight LDS-DMA to vmcnt(0) every tile. 48 KiB LDS ->
-// 3 blocks/CU. Targets the low-occupancy long-K shapes (fc2 class).
+// NBUF-buffer glds pipeline (K % 64 == 0): NBUF-2 K-tiles of loads stay in
+// flight ACROSS barriers (guide "Pipelining across barriers"): raw
+// s_barrier + counted `s_waitcnt vmcnt(N)` instead of __syncthreads(),
+// whose fence would drain the in-flight LDS-DMA to vmcnt(0) every tile.
+// NBUF=3: 48 KiB LDS -> 3 blocks/CU (the default t7). NBUF=4 (t10): one
+// more tile in flight at 64 KiB -> 2 blocks/CU - a depth-vs-occupancy
+// trade for the latency-bound regime the final PMC shows
+// (profiles/hot_pmc_final.csv: MemUnitStalled ~0.1, zero LDS conflicts).
 // ---------------------------------------------------------------------------
-template <int MI, int NI, bool XCD = false, int EPM = 0>
+template <int MI, int NI, bool XCD = false, int EPM = 0, int NBUF = 3>
 __global__ __launch_bounds__(256) void gemm_bf16_glds3_kernel(
         const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
     constexpr int TBM = 32 * MI;
     constexpr int TBN = 32 * NI;
     constexpr int KB = BK;
-    __shared__ __bf16 smem[3][TBM + TBN][KB];
+    __shared__ __bf16 smem[NBUF][TBM + TBN][KB];
 
     int bx = blockIdx.x, by = blockIdx.y;
     if (XCD) {
@@ -331,20 +334,31 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds3_kernel(
 
     f32x4 acc[MI][NI] = {};
     const int n_tiles = K / KB;
-    stage(0, 0);
-    if (n_tiles > 1) stage(1, KB);
+    #pragma unroll
+    for (int p = 0; p < NBUF - 1; ++p)
+        if (p < n_tiles) stage(p, p * KB);
 
     for (int t = 0; t < n_tiles; ++t) {
-        // tile t's LDS-DMA retired for THIS wave; up to one more tile stays
-        // in flight across the barrier
-        if (t + 1 < n_tiles)
+        // tile t's LDS-DMA retired for THIS wave; up to NBUF-2 more tiles
+        // stay in flight across the barrier
+        const int inflight = min(NBUF - 2, n_tiles - 1 - t);
+        switch (inflight) {
+        case 0:
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            break;
+        case 1:
             asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GLDS_PER_TILE)
                          : "memory");
-        else
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            break;
+        default:
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GLDS_PER_TILE)
+                         : "memory");
+            break;
+        }
         __builtin_amdgcn_s_barrier();
-        if (t + 2 < n_tiles) stage((t + 2) % 3, (t + 2) * KB);
-        const int buf = t % 3;
+        if (t + NBUF - 1 < n_tiles)
+            stage((t + NBUF - 1) % NBUF, (t + NBUF - 1) * KB);
+        const int buf = t % NBUF;
         #pragma unroll
         for (int kk = 0; kk < KB; kk += 32) {
             const int kgrp = kk + (lane >> 4) * 8;
@@ -564,33 +578,37 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
     } while (0)
     if (K % BK == 0) {
-#define PE_LAUNCH_GLDS3(MI_, NI_, GRID_)                                     \
+#define PE_LAUNCH_GLDS3(MI_, NI_, NBUF_, GRID_)                              \
     do {                                                                     \
         if (epm == 1)                                                        \
-            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 1>),  \
-                               GRID_, dim3(256), 0, pe_current_stream(),     \
-                               ap, bp, cp, M, N, K, ep);                     \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds3_kernel<MI_, NI_, true, 1, NBUF_>), GRID_,   \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
         else if (epm == 2)                                                   \
-            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 2>),  \
-                               GRID_, dim3(256), 0, pe_current_stream(),     \
-                               ap, bp, cp, M, N, K, ep);                     \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds3_kernel<MI_, NI_, true, 2, NBUF_>), GRID_,   \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
         else                                                                 \
-            hipLaunchKernelGGL((gemm_bf16_glds3_kernel<MI_, NI_, true, 0>),  \
-                               GRID_, dim3(256), 0, pe_current_stream(),     \
-                               ap, bp, cp, M, N, K, ep);                     \
+            hipLaunchKernelGGL(                                              \
+                (gemm_bf16_glds3_kernel<MI_, NI_, true, 0, NBUF_>), GRID_,   \
+                dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
     } while (0)
         if (t == 7) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-            PE_LAUNCH_GLDS3(2, 2, grid);
+            PE_LAUNCH_GLDS3(2, 2, 3, grid);
+        } else if (t == 10) {
+            // 4-buffer pipeline: one more tile in flight, 2 blocks/CU
+            dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+            PE_LAUNCH_GLDS3(2, 2, 4, grid);
         } else if (t == 8) {
             // 128x64 3-buffer: double arithmetic intensity vs t7 (the 64x64
             // tiles are HBM/L2-bound at ~32 FLOP/B); 72 KiB LDS -> 2/CU
             dim3 grid((M + 127) / 128, (N + BN - 1) / BN);
-            PE_LAUNCH_GLDS3(4, 2, grid);
+            PE_LAUNCH_GLDS3(4, 2, 3, grid);
         } else if (t == 9) {
             // 128x128 3-buffer: 96 KiB LDS -> 1/CU
             dim3 grid((M + 127) / 128, (N + 127) / 128);
-            PE_LAUNCH_GLDS3(4, 4, grid);
+            PE_LAUNCH_GLDS3(4, 4, 3, grid);
         } else if (t == 4 || t == 5 || t == 6) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             if (t == 4) PE_LAUNCH_GLDS(2, 2, 2, false, grid);

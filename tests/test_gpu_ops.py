@@ -50,7 +50,7 @@ def test_gemm_matches_fp32(m, n, k):
     _assert_close(out.cpu(), ref, atol=0.05 * k ** 0.5 * 0.05)
 
 
-@pytest.mark.parametrize("tile", [1, 2, 3, 4, 5, 6, 7, 8, 9])
+@pytest.mark.parametrize("tile", [1, 2, 3, 4, 5, 6, 7, 8, 9, 10])
 def test_gemm_explicit_tiles(tile):
     """Every dispatchable tile variant (incl. the 3-buffer 128x64/128x128
     glds3 instantiations) against the fp32 reference on a K%64==0 shape."""

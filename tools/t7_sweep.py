@@ -21,7 +21,7 @@ shapes = [
     ("bert-qkv", 1024, 2304, 768),
     ("bert-fc2", 1024, 768, 3072),
 ]
-TILES = (5, 7, 8, 9)
+TILES = (5, 7, 10)
 
 def bench(fn, iters=60):
     for _ in range(12):
