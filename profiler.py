@@ -102,10 +102,26 @@ def profile_module_shard(module_cfg: dict, inputs, layer_start: int,
     }
 
 
+def _np_pack(data):
+    """Tensors -> by-value numpy for the mp queue: torch's shared-storage
+    tensor pickling needs the CHILD alive while the parent materializes the
+    storage, and the child exits right after put() (ConnectionResetError)."""
+    if isinstance(data, tuple):
+        return tuple(t.detach().float().cpu().numpy() for t in data)
+    return data.detach().float().cpu().numpy()
+
+
+def _np_unpack(data):
+    if isinstance(data, tuple):
+        return tuple(torch.from_numpy(a) for a in data)
+    return torch.from_numpy(data)
+
+
 def _profile_mp_queue(queue, module_cfg, inputs, layer_start, layer_end,
                       warmup, iterations):
     res = profile_module_shard(module_cfg, inputs, layer_start, layer_end,
                                warmup, iterations)
+    res["_outputs"] = _np_pack(res["_outputs"])
     queue.put(res)
 
 
@@ -121,6 +137,7 @@ def profile_module_shard_mp(module_cfg, inputs, layer_start, layer_end,
     proc.start()
     result = queue.get()
     proc.join()
+    result["_outputs"] = _np_unpack(result["_outputs"])
     return result
 
 

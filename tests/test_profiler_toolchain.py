@@ -118,3 +118,21 @@ def test_save_model_weights_cli(tmp_path):
                     "-M", str(out)], check=True,
                    env=dict(os.environ, PYTHONPATH=REPO), cwd=REPO)
     assert out.exists()
+
+
+def test_profiler_cli_subprocess_mode(tmp_path):
+    """The real profiler CLI over a few layers - exercises the fresh-
+    subprocess measurement path (regression: torch shared-storage tensors
+    on the mp queue raced the child's exit -> ConnectionResetError)."""
+    env = dict(os.environ, PYTHONPATH=REPO)
+    out_yml = tmp_path / "profiler_results.yml"
+    subprocess.run(
+        [sys.executable, os.path.join(REPO, "profiler.py"),
+         "-m", "google/vit-base-patch16-224", "-b", "2", "-i", "1",
+         "-w", "0", "-l", "1", "-L", "3", "-o", str(out_yml)],
+        check=True, env=env, cwd=str(tmp_path), timeout=600)
+    with open(out_yml) as f:
+        res = yaml.safe_load(f)
+    assert res["layers"] == 48 and len(res["profile_data"]) == 3
+    assert res["profile_data"][0]["shape_out"] == [[197, 768], [197, 768]]
+    assert all(d["time"] > 0 for d in res["profile_data"])
