@@ -303,3 +303,38 @@ def test_vit_huge_forward_hip_vs_cpu_fp32():
     cos = torch.nn.functional.cosine_similarity(
         out.float().cpu().flatten(), ref.flatten(), dim=0)
     assert cos.item() > 0.99, cos.item()
+
+
+def test_folded_ln_linear_matches_reference():
+    """The gated-off folded-LN path (ops.ln_linear + make_ln_fold + the
+    EPM=1 GEMM epilogue + ln_stats kernel) against LN->linear fp32."""
+    torch.manual_seed(7)
+    m, n, k = 197, 768, 768
+    x = torch.randn(m, k) * 0.7 + 0.2
+    lw = torch.randn(k) * 0.3 + 1.0
+    lb = torch.randn(k) * 0.1
+    w = torch.randn(n, k) * 0.03
+    b = torch.randn(n) * 0.1
+    ref = torch_ref.linear(torch_ref.layernorm(x, lw, lb, 1e-6), w, b)
+    xg = x.bfloat16().cuda()
+    fused = ops.make_ln_fold(w.bfloat16().cuda(), lw.bfloat16().cuda(),
+                             lb.bfloat16().cuda())
+    out = ops.ln_linear(xg, lw.bfloat16().cuda(), lb.bfloat16().cuda(), 1e-6,
+                        w.bfloat16().cuda(), b.bfloat16().cuda(), fused=fused)
+    _assert_close(out.cpu(), ref, atol=0.25)
+
+
+def test_gemm_want_stats_emits_row_stats():
+    """EPM=2: the GEMM's per-row (sum, sumsq) partials reduce to the true
+    row statistics of its output."""
+    torch.manual_seed(8)
+    m, n, k = 197, 768, 768
+    a = torch.randn(m, k) * 0.5
+    w = torch.randn(n, k) * 0.03
+    out, stats = _hip().linear(a.bfloat16().cuda(), w.bfloat16().cuda(),
+                               want_stats=True)
+    assert stats is not None and stats.shape[0] == m and stats.shape[2] == 2
+    s = stats.float().sum(dim=1).cpu()  # reduce partials -> [m, 2]
+    of = out.float().cpu()
+    assert torch.allclose(s[:, 0], of.sum(dim=1), rtol=2e-2, atol=1.0)
+    assert torch.allclose(s[:, 1], (of * of).sum(dim=1), rtol=2e-2, atol=2.0)
