@@ -6,7 +6,6 @@ the (public) architecture hyperparameters for the supported model zoo so the
 framework is fully offline.
 """
 import dataclasses
-from typing import Optional
 
 
 @dataclasses.dataclass

@@ -20,8 +20,8 @@ from torch import Tensor, nn
 from .. import ops
 from . import ModuleShard, ModuleShardConfig
 from .config import TransformerConfig
-from .vit import (ViTLayerShard, ViTModelShard, ViTShardForImageClassification,
-                  _param, _t, init_param, load_npz_or_random)
+from .vit import (ViTLayerShard, ViTModelShard,
+                  ViTShardForImageClassification, _param, _t, init_param)
 
 logger = logging.getLogger(__name__)
 

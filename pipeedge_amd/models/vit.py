@@ -25,7 +25,7 @@ import logging
 import math
 import zlib
 from collections.abc import Mapping
-from typing import Optional, Union
+from typing import Union
 
 import numpy as np
 import torch
