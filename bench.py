@@ -49,9 +49,22 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     import torch.distributed as dist
 
     use_cuda = torch.cuda.is_available()
+    backend = os.environ.get("PIPEEDGE_BENCH_BACKEND",
+                             "nccl" if use_cuda else "gloo")
     if use_cuda:
-        # modulo lets N ranks share fewer GPUs (single-GPU RCCL validation)
-        local_rank = int(os.environ.get("LOCAL_RANK", rank))             % torch.cuda.device_count()
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        n_dev = torch.cuda.device_count()
+        if local_rank >= n_dev:
+            if backend == "nccl":
+                # NCCL/RCCL forbids two ranks on one device ("Duplicate GPU
+                # detected") - fail loud instead of hanging in init
+                raise SystemExit(
+                    f"bench: rank {rank} needs cuda:{local_rank} but only "
+                    f"{n_dev} GPU(s) are visible; RCCL requires one GPU per "
+                    f"rank. Use --gpus <= device_count, or set "
+                    f"PIPEEDGE_BENCH_BACKEND=gloo to validate the pipeline "
+                    f"logic with shared GPUs (CPU-staged wire).")
+            local_rank %= n_dev  # gloo tolerates device sharing
         torch.cuda.set_device(local_rank)
         devices.initialize(f"cuda:{local_rank}")
     else:
@@ -62,7 +75,6 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     distributed = world_size > 1
     results_group = None
     if distributed:
-        backend = "nccl" if use_cuda else "gloo"
         dist.init_process_group(backend, rank=rank, world_size=world_size)
         # dedicated communicator for the results back-edge (see
         # DistP2pPipelineStage group docs)

@@ -273,6 +273,14 @@ class DistP2pPipelineStage:
         self.device = torch.device(device) if device is not None else \
             (torch.device("cuda", torch.cuda.current_device())
              if torch.cuda.is_available() else torch.device("cpu"))
+        # Wire device: where tensors live for dist send/recv. RCCL moves
+        # GPU-resident buffers directly; gloo requires CPU staging, so a
+        # gloo data plane with GPU compute (e.g. multiple stages sharing
+        # one GPU for validation) stages through pinned-less CPU copies.
+        self.wire_device = self.device
+        if self.device.type == "cuda" and dist.is_initialized() and \
+                dist.get_backend() == "gloo":
+            self.wire_device = torch.device("cpu")
         # transforms (quantization hooks attach here)
         self.recv_transform = _default_recv_transform
         self.send_transform = _default_send_transform
@@ -324,12 +332,22 @@ class DistP2pPipelineStage:
         return torch.cuda.stream(streams[i % len(streams)])
 
     # -- stage-rank batch processing -----------------------------------------
+    def _from_wire(self, tensors: List[torch.Tensor]) -> List[torch.Tensor]:
+        if self.wire_device == self.device:
+            return tensors
+        return [t.to(self.device) for t in tensors]
+
+    def _to_wire(self, tensors: List[torch.Tensor]) -> List[torch.Tensor]:
+        if self.wire_device == self.device:
+            return tensors
+        return [t.to(self.wire_device) for t in tensors]
+
     def stage_run_batch(self, n_ubatch: int):
         """Process one announced batch: exactly ``n_ubatch`` recv->work->send."""
         assert self.work_cb is not None
         streams = self._ustreams()
         depth = max(2, len(streams) if streams else 2)
-        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device,
+        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.wire_device,
                              depth=depth, group=self.recv_group) \
             if self.rank_src is not None else None
         send = _SendWindow(self.rank_dst, group=self.send_group) \
@@ -339,12 +357,12 @@ class DistP2pPipelineStage:
         for i in range(n_ubatch):
             with self._stream_ctx(streams, i):
                 self._call_hook(self._recv_pre_hook)
-                wire_in = recv.wait(i)
+                wire_in = self._from_wire(recv.wait(i))
                 self._call_hook(self._recv_post_hook, wire_in)
                 inputs = self.recv_transform(wire_in)
                 outputs = self.work_cb(inputs)
                 recv.mark_consumed(i)
-                wire_out = self.send_transform(outputs)
+                wire_out = self._to_wire(self.send_transform(outputs))
                 self._call_hook(self._send_pre_hook)
                 send.send(wire_out)
                 self._call_hook(self._send_post_hook, wire_out)
@@ -371,7 +389,7 @@ class DistP2pPipelineStage:
                     s.synchronize()
             return
         send = _SendWindow(self.rank_dst, group=self.send_group)
-        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.device,
+        recv = _RecvPipeline(self.rank_src, self.recv_specs, self.wire_device,
                              group=self.recv_group)
         streams = self._ustreams()
         recv.post(2, n)
@@ -380,9 +398,10 @@ class DistP2pPipelineStage:
             with self._stream_ctx(streams, i):
                 if self.work_cb is not None:
                     outputs = self.work_cb(self._to_device(ub))
-                    wire = self.send_transform(outputs)
+                    wire = self._to_wire(self.send_transform(outputs))
                 else:
-                    wire = _default_send_transform(self._to_device(ub))
+                    wire = self._to_wire(
+                        _default_send_transform(self._to_device(ub)))
                 self._call_hook(self._send_pre_hook)
                 send.send(wire)
                 self._call_hook(self._send_post_hook, wire)

@@ -59,11 +59,12 @@ __global__ __launch_bounds__(256) void minmax_kernel(
 __global__ __launch_bounds__(256) void pack_kernel(
         const __bf16* __restrict__ x, unsigned int* __restrict__ words,
         const float* __restrict__ mn, const float* __restrict__ mx,
-        long per_image, long words_per_image, long pad_words, int bit) {
+        long n_total, long per_image, long words_per_image, long pad_words,
+        int bit) {
     const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (gid >= n_total) return;  // grid rounds up to blockDim multiples
     const long img = gid / pad_words;
     const long w = gid % pad_words;
-    if (w >= pad_words) return;
     if (w >= words_per_image) {  // padding for the static wire schema
         words[gid] = 0u;
         return;
@@ -73,7 +74,6 @@ __global__ __launch_bounds__(256) void pack_kernel(
     const float scale_raw = mx[img] - shift;
     const float scale = scale_raw != 0.0f ? scale_raw : 1.0f;
     const double levels = (double)((1ull << bit) - 1ull);
-    const long base = img * per_image + w * ratio;
     unsigned int word = 0;
     for (int i = 0; i < ratio; ++i) {
         const long idx = w * (long)ratio + i;
@@ -87,15 +87,15 @@ __global__ __launch_bounds__(256) void pack_kernel(
         }
     }
     words[gid] = word;
-    (void)base;
 }
 
 // one thread per packed word; writes ratio bf16 values
 __global__ __launch_bounds__(256) void unpack_kernel(
         const unsigned int* __restrict__ words, __bf16* __restrict__ out,
         const float* __restrict__ scale, const float* __restrict__ shift,
-        long per_image, long pad_words, int bit) {
+        long n_total, long per_image, long pad_words, int bit) {
     const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (gid >= n_total) return;  // grid rounds up to blockDim multiples
     const long img = gid / pad_words;
     const long w = gid % pad_words;
     const int ratio = 32 / bit;
@@ -141,8 +141,8 @@ std::vector<torch::Tensor> quant_encode_bf16(torch::Tensor x, int64_t bit,
                        stream,
                        reinterpret_cast<const __bf16*>(x_c.data_ptr()),
                        reinterpret_cast<unsigned int*>(comm.data_ptr()),
-                       mn.data_ptr<float>(), mx.data_ptr<float>(), per_image,
-                       words_per_image, padw, (int)bit);
+                       mn.data_ptr<float>(), mx.data_ptr<float>(), total,
+                       per_image, words_per_image, padw, (int)bit);
     HIP_CHECK_LAST();
     auto scale = (mx - mn);
     return {comm, scale, mn};
@@ -168,7 +168,7 @@ torch::Tensor quant_decode_bf16(torch::Tensor comm, torch::Tensor scale,
                        reinterpret_cast<const unsigned int*>(comm_c.data_ptr()),
                        reinterpret_cast<__bf16*>(out.data_ptr()),
                        scale_f.data_ptr<float>(), shift_f.data_ptr<float>(),
-                       per_image, padw, (int)bit);
+                       total, per_image, padw, (int)bit);
     HIP_CHECK_LAST();
     return out;
 }

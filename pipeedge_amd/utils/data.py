@@ -31,6 +31,22 @@ class RolloverTensorDataset(Dataset):
         return tuple(t[base] for t in self.tensors)
 
 
+class ShuffledDataset(Dataset):
+    """Index-permuted view of a dataset (``--dataset-shuffle``; the reference
+    shuffles at load time, runtime.py:358-401)."""
+
+    def __init__(self, base: Dataset, seed: int = 0):
+        self.base = base
+        self.perm = torch.randperm(
+            len(base), generator=torch.Generator().manual_seed(seed)).tolist()
+
+    def __len__(self):
+        return len(self.base)
+
+    def __getitem__(self, idx):
+        return self.base[self.perm[idx]]
+
+
 def load_dataset_synthetic(model_config, batch_size: int,
                            seq_len: Optional[int] = None,
                            seed: int = 0) -> Dataset:

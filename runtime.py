@@ -386,17 +386,28 @@ def load_dataset(dataset_cfg: dict, model_name: str, batch_size: int,
     random inputs (offline environment)."""
     config = model_cfg.get_model_config(model_name)
     name = dataset_cfg.get("name")
+    indices = dataset_cfg.get("indices")
     if name == "CoLA":
         from pipeedge_amd.utils.data import load_dataset_glue
-        return load_dataset_glue(model_name, "cola", dataset_cfg.get("split", "train"),
-                                 batch_size, indices=dataset_cfg.get("indices"))
-    if name == "ImageNet":
+        dataset = load_dataset_glue(model_name, "cola",
+                                    dataset_cfg.get("split", "train"),
+                                    batch_size, indices=indices)
+    elif name == "ImageNet":
         from pipeedge_amd.utils.data import load_dataset_imagenet
-        return load_dataset_imagenet(dataset_cfg["root"],
-                                     dataset_cfg.get("split", "train"),
-                                     batch_size,
-                                     indices=dataset_cfg.get("indices"))
-    return load_dataset_synthetic(config, batch_size)
+        dataset = load_dataset_imagenet(dataset_cfg["root"],
+                                        dataset_cfg.get("split", "train"),
+                                        batch_size, indices=indices)
+    else:
+        dataset = load_dataset_synthetic(config, batch_size)
+        if indices is not None:
+            from pipeedge_amd.utils.data import RolloverTensorDataset
+            sel = [dataset[int(i) % len(dataset)] for i in indices]
+            dataset = RolloverTensorDataset(
+                batch_size, *(torch.stack(ts) for ts in zip(*sel)))
+    if dataset_cfg.get("shuffle"):
+        from pipeedge_amd.utils.data import ShuffledDataset
+        dataset = ShuffledDataset(dataset)
+    return dataset
 
 
 def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
@@ -499,6 +510,17 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
                              for s in out_specs]
                 stage_ctx.send_transform = make_quant_send_transform(
                     module, pad_words, stage_quant[stage])
+            elif stage_layers[stage][1] < total_layers and \
+                    os.getenv(ENV_ADAPTIVE_QUANT):
+                # static wire schema: a stage configured quant=0 has no quant
+                # send transform, so adaptive policies cannot enable
+                # quantization for it mid-run (the reference could; its wire
+                # re-negotiates per message). Surface it instead of no-op'ing.
+                logger.warning(
+                    "%s is set but stage %d has quant=0: adaptive "
+                    "quantization is DISABLED for this stage (configure a "
+                    "nonzero -q ceiling to enable; policy bitwidth 0 is "
+                    "coerced to that ceiling)", ENV_ADAPTIVE_QUANT, stage)
             # inbound: dequantize unless we're stage 0
             if stage > 0 and stage_quant[stage - 1] > 0:
                 stage_ctx.recv_transform = make_quant_recv_transform(dtype)
