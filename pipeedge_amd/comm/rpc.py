@@ -118,10 +118,17 @@ class DistRpcPipelineStage:
             with self._sem_mod:
                 device = next(self._module.parameters(), torch.empty(0)).device
                 dtype = getattr(self._module, "compute_dtype", None)
+                # a quantized wire (5-tensor groups headed by a uint8 comm
+                # tensor) must keep its f32 scale/shift exactly - the decode
+                # pre-hook casts AFTER dequantization
+                is_wire = (isinstance(inputs, tuple) and len(inputs) >= 5
+                           and len(inputs) % 5 == 0
+                           and inputs[0].dtype == torch.uint8)
 
                 def to_dev(t):
                     t = t.to(device)
-                    if dtype is not None and t.is_floating_point():
+                    if not is_wire and dtype is not None \
+                            and t.is_floating_point():
                         t = t.to(dtype)
                     return t
                 if isinstance(inputs, torch.Tensor):
@@ -205,6 +212,12 @@ def run_pipeline_rpc(world_size: int, rank: int, model_name: str,
                          acc_type="layers")
     monitoring_glue.add_key(runtime_mod.MONITORING_KEY_OUTPUT,
                             work_type="classifications", acc_type="correct")
+    monitoring_glue.add_key(runtime_mod.MONITORING_KEY_QUANT_DECODE,
+                            work_type="tensors", acc_type="bits")
+    monitoring_glue.add_key(runtime_mod.MONITORING_KEY_QUANT_ENCODE,
+                            work_type="tensors", acc_type="bits")
+    monitoring_glue.add_key(runtime_mod.MONITORING_KEY_RECV, work_type="Mbits")
+    monitoring_glue.add_key(runtime_mod.MONITORING_KEY_SEND, work_type="Mbits")
     opts = tensorpipe_rpc_backend_options_factory(
         num_worker_threads=rpc_num_worker_threads)
     with DistRpcContext(world_size, rank,
@@ -219,6 +232,34 @@ def run_pipeline_rpc(world_size: int, rank: int, model_name: str,
             pipeline = model_cfg.dist_rpc_pipeline_factory(
                 model_name, model_file, stage_ranks, stage_layers, rank,
                 runtime_mod.handle_results)
+            # quant + monitoring wiring (reference runtime.py:558-565):
+            # per-stage quant_bit / rate_constraint buffers, monitor hooks on
+            # every stage, quant encode on all-but-last, decode on
+            # all-but-first
+            import os
+            send_constraint = float(
+                os.getenv(runtime_mod.ENV_SEND_CONSTRAINT, "0"))
+            pipeline.rpc_register_buffer(
+                "quant_bit", [torch.tensor(q) for q in stage_quant],
+                persistent=False)
+            pipeline.rpc_register_buffer(
+                "rate_constraint",
+                [torch.tensor(send_constraint) for _ in stage_quant],
+                persistent=False)
+            pipeline.rpc_register_forward_hook(
+                runtime_mod.forward_hook_monitor)
+            pipeline.rpc_register_forward_hook(
+                runtime_mod.forward_hook_quant_encode, last=False)
+            # send/recv monitors see wire sizes: send AFTER encode, recv
+            # BEFORE decode (torch runs hooks in registration order)
+            pipeline.rpc_register_forward_hook(
+                runtime_mod.forward_hook_rpc_send_monitor, last=False)
+            pipeline.rpc_register_forward_pre_hook(
+                runtime_mod.forward_pre_hook_rpc_recv_monitor, first=False)
+            pipeline.rpc_register_forward_pre_hook(
+                runtime_mod.forward_pre_hook_quant_decode, first=False)
+            pipeline.rpc_register_forward_pre_hook(
+                runtime_mod.forward_pre_hook_monitor)
             dataset = runtime_mod.load_dataset(dataset_cfg, model_name,
                                                batch_size, ubatch_size)
             start = runtime_mod.results_counter.value

@@ -239,7 +239,12 @@ def set_quant_controller(module, outputs) -> None:
         bw2 = module.bitwidth2.item()
         bw1_iters = module.bitwidth1_iters.item()
     except AttributeError:
-        bw1 = bw2 = bw1_iters = 0
+        # first window period, before the controller sets these buffers:
+        # hold the CONFIGURED bitwidth (the reference zeroes quant_bit here,
+        # reference runtime.py:183-191, disabling quantization for the first
+        # window even when -q was given; we keep the configured bit instead)
+        bw1 = bw2 = module.quant_bit.item()
+        bw1_iters = 0
     with monitoring.get_locked_context(MONITORING_KEY_SEND) as mctx:
         tag = mctx.get_tag(key=MONITORING_KEY_SEND)
         window_size = mctx.get_window_size(key=MONITORING_KEY_SEND)
@@ -328,6 +333,98 @@ def make_quant_recv_transform(dtype: torch.dtype):
         outputs = tensors[0] if len(tensors) == 1 else tuple(tensors)
         return outputs
     return transform
+
+
+# ---------------------------------------------------------------------------
+# RPC-path quantization/monitoring forward hooks (parity: reference
+# runtime.py:60-119,558-565). The p2p engine attaches the same logic as
+# stage send/recv TRANSFORMS (above); the RPC transport owns nn.Modules
+# remotely, so there it rides torch's hook mechanism, fanned out via
+# DistRpcPipeline.rpc_register_*. Unlike the p2p wire, TensorPipe frames are
+# dynamic, so adaptive bitwidths are NOT clamped to a schema ceiling here -
+# bit 0 (quantization off) passes through unquantized like the reference.
+# ---------------------------------------------------------------------------
+
+def forward_pre_hook_monitor(_module, _inputs) -> None:
+    """Mark shard-forward iteration start (reference runtime.py:60-62)."""
+    monitoring.iteration_start(MONITORING_KEY_MODEL)
+
+
+def forward_hook_monitor(module, _inputs, outputs) -> None:
+    """Record shard-forward completion (reference runtime.py:64-71)."""
+    n_items = models.get_microbatch_size(outputs, verify=True)
+    n_layers = module.shard_config.layer_end - module.shard_config.layer_start + 1
+    monitoring.iteration(MONITORING_KEY_MODEL, work=n_items, accuracy=n_layers)
+
+
+def forward_hook_quant_encode(module, _inputs, outputs):
+    """Quantize stage outputs into the 5-tensor wire format (reference
+    runtime.py:73-91), applying the adaptive policy first when configured."""
+    policy = _ADAPTIVE_POLICIES.get(os.getenv(ENV_ADAPTIVE_QUANT, ""))
+    if policy is not None:
+        try:
+            policy(module, outputs)
+        except Exception:  # noqa: BLE001
+            logger.exception("adaptive quant policy failed")
+    monitoring.iteration_start(MONITORING_KEY_QUANT_ENCODE)
+    if isinstance(outputs, torch.Tensor):
+        outputs = (outputs,)
+    quant_bit = int(module.quant_bit.item())
+    wire: List[torch.Tensor] = []
+    for tensor in outputs:
+        if quant_bit > 0:
+            clamp = clamp_banner2019_laplace if tensor.min() < 0.2 \
+                else clamp_banner2019_gelu
+            tensor = clamp(tensor, quant_bit)
+        wire += tensor_encode_outerdim(tensor, quant_bit)
+    n_items = models.get_microbatch_size(outputs[0]) if quant_bit > 0 else 0
+    monitoring.iteration(MONITORING_KEY_QUANT_ENCODE, work=n_items,
+                         accuracy=quant_bit)
+    return tuple(wire)
+
+
+def forward_hook_rpc_send_monitor(_module, _inputs, outputs) -> None:
+    """Record outbound wire traffic on the RPC path. The p2p engine measures
+    this in its comm-thread hooks (p2p_*_hook_monitor below); TensorPipe
+    pushes happen inside the remote stage call, so the wire size is sampled
+    from the (post-encode) outputs instead. The adaptive policies key off
+    this MONITORING_KEY_SEND window state."""
+    tensors = (outputs,) if isinstance(outputs, torch.Tensor) else outputs
+    n_bytes = sum(t.numel() * t.element_size() for t in tensors)
+    monitoring.iteration_start(MONITORING_KEY_SEND)
+    monitoring.iteration(MONITORING_KEY_SEND, work=n_bytes * 8 / 1e6)
+
+
+def forward_pre_hook_rpc_recv_monitor(_module, input_arg) -> None:
+    """Record inbound wire traffic on the RPC path (mirror of the send
+    monitor; registered before the decode pre-hook so it sees wire sizes)."""
+    wire = input_arg[0]
+    tensors = (wire,) if isinstance(wire, torch.Tensor) else wire
+    n_bytes = sum(t.numel() * t.element_size() for t in tensors)
+    monitoring.iteration_start(MONITORING_KEY_RECV)
+    monitoring.iteration(MONITORING_KEY_RECV, work=n_bytes * 8 / 1e6)
+
+
+def forward_pre_hook_quant_decode(module, input_arg):
+    """Decode 5-tensor wire groups back into stage inputs (reference
+    runtime.py:93-119). ``input_arg`` is the 1-tuple of forward() args."""
+    monitoring.iteration_start(MONITORING_KEY_QUANT_DECODE)
+    assert isinstance(input_arg, tuple) and len(input_arg) == 1
+    wire = input_arg[0]
+    assert len(wire) % 5 == 0 and len(wire) >= 5
+    quant_bit = int(wire[4].view(-1)[0].item())
+    dtype = getattr(module, "compute_dtype", None)
+    tensors = []
+    for i in range(len(wire) // 5):
+        decoded = tensor_decode_outerdim(list(wire[i * 5:i * 5 + 5]))
+        if dtype is not None:
+            decoded = decoded.to(dtype)
+        tensors.append(decoded)
+    outputs = tuple(tensors) if len(tensors) == 1 else (tuple(tensors),)
+    n_items = models.get_microbatch_size(tensors[0]) if quant_bit > 0 else 0
+    monitoring.iteration(MONITORING_KEY_QUANT_DECODE, work=n_items,
+                         accuracy=quant_bit)
+    return outputs
 
 
 def cast_recv_transform(dtype: torch.dtype):

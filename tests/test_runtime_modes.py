@@ -62,6 +62,51 @@ def test_runtime_adaptive_quant_controller(tmp_path):
     assert "Adaptive quantization (controller)" in outs[0][1], outs[0][1]
 
 
+def test_runtime_rpc_quantized_adaptive(tmp_path):
+    """RPC-path quant parity (reference runtime.py:558-565): `-c rpc -q 8,0`
+    must quantize the stage-0 -> stage-1 hop and run the adaptive policy."""
+    env = {"ADAPTIVE_QUANT": "CONTROLLER", "SEND_CONSTRAINT": "100",
+           "WINDOW_SIZE": "2"}
+    # per-rank cwds: each rank writes its own monitoring CSVs
+    cwds = []
+    for r in range(2):
+        d = tmp_path / f"rank{r}"
+        d.mkdir()
+        cwds.append(str(d))
+    ctx = mp.get_context("spawn")
+    result_q = ctx.Queue()
+    port = 29641
+    extra = ["-c", "rpc", "-b", "8", "-u", "2", "-m", MODEL,
+             "-pt", "1,24,25,48", "-q", "8,0"]
+    procs = [ctx.Process(target=_worker,
+                         args=(r, port, extra, env, result_q, cwds[r]))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(2):
+        rank, rc, out, err = result_q.get(timeout=300)
+        outs[rank] = (rc, out, err)
+    for p in procs:
+        p.join(timeout=30)
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]
+    # quantized transport: the encode hook ran on the stage-0 owner (rank 0)
+    # and recorded per-iteration bitwidths in its monitoring CSV
+    enc = (tmp_path / "rank0" / "quant_encode.csv").read_text() \
+        .strip().splitlines()
+    assert len(enc) > 1, enc
+    bits = {int(float(row.split(",")[-1])) for row in enc[1:]}
+    assert bits - {0}, f"no nonzero quant bitwidths recorded: {enc}"
+    # decode fired on the stage-1 owner (rank 1)
+    dec = (tmp_path / "rank1" / "quant_decode.csv").read_text() \
+        .strip().splitlines()
+    assert len(dec) > 1, dec
+    # the adaptive controller fired at the window boundary
+    assert "Adaptive quantization (controller)" in outs[0][1], outs[0][1]
+
+
 def test_runtime_auto_sched_from_yaml(tmp_path):
     layers = 48
     model = {"layers": layers, "parameters_in": 150528,
