@@ -98,12 +98,13 @@ class BertLayerShard(ModuleShard):
             init_param(p, f"{prefix}/block{self.block_id}/{name}", kind)
 
     @torch.no_grad()
-    def forward(self, data):
+    def forward(self, data, attn_mask=None):
         cfg = self.config
         eps = cfg.layer_norm_eps
         if self.has_layer(0):
             qkv = ops.linear(data, self.qkv_weight, self.qkv_bias)
-            ctx = ops.qkv_attention(qkv, cfg.num_attention_heads, self.scale)
+            ctx = ops.qkv_attention(qkv, cfg.num_attention_heads, self.scale,
+                                    attn_mask=attn_mask)
             data = (ctx, data)
         if self.has_layer(1):
             y = ops.linear(data[0], self.out_weight, self.out_bias)
@@ -191,15 +192,37 @@ class BertModelShard(ModuleShard):
 
     @torch.no_grad()
     def forward(self, data):
+        """Forward with padding-mask plumbing.
+
+        The first shard derives the additive attention mask from the token
+        ids (pad_token_id positions -> -1e9; the reference's shards run
+        MASK-FREE over padding, reference bert.py:41-52 - this build fixes
+        that) and every stage boundary carries the mask as its LAST payload
+        tensor (spec.boundary_spec appends it) so downstream attention
+        sublayers mask correctly; the last shard drops it.
+        """
+        attn_mask = None
         if self.shard_config.is_first:
-            data = self.embeddings(data)
+            ids = data
+            pad = self.config.pad_token_id
+            dt = self.embeddings.word_embeddings.dtype
+            attn_mask = torch.where(
+                ids == pad,
+                torch.full((), -1e9, dtype=dt, device=ids.device),
+                torch.zeros((), dtype=dt, device=ids.device))
+            data = self.embeddings(ids)
+        elif isinstance(data, tuple):
+            *inner, attn_mask = data
+            data = inner[0] if len(inner) == 1 else tuple(inner)
         for layer in self.layers:
-            data = layer(data)
+            data = layer(data, attn_mask)
         if self.shard_config.is_last:
             # pooler: tanh(dense(CLS token))
-            data = ops.linear(data[:, 0], self.pooler_weight, self.pooler_bias,
-                              act="tanh")
-        return data
+            return ops.linear(data[:, 0], self.pooler_weight,
+                              self.pooler_bias, act="tanh")
+        if isinstance(data, torch.Tensor):
+            return (data, attn_mask)
+        return (*data, attn_mask)
 
     @staticmethod
     def save_weights(model_name: str, model_file: str, seed: int = 0) -> None:

@@ -28,6 +28,7 @@ class TransformerConfig:
     vocab_size: int = 0
     max_position_embeddings: int = 0
     type_vocab_size: int = 0
+    pad_token_id: int = 0           # BERT [PAD]; masks attention over padding
 
     @property
     def head_dim(self) -> int:

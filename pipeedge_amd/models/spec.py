@@ -41,12 +41,18 @@ def boundary_spec(config: TransformerConfig, layer_end: int, ubatch_size: int,
     i = config.intermediate_size
     sub = (layer_end - 1) % 4
     if sub == 0:
-        return [TensorSpec((ubatch_size, s, d), dtype),
-                TensorSpec((ubatch_size, s, d), dtype)]
-    if sub == 2:
-        return [TensorSpec((ubatch_size, s, i), dtype),
-                TensorSpec((ubatch_size, s, d), dtype)]
-    return [TensorSpec((ubatch_size, s, d), dtype)]
+        specs = [TensorSpec((ubatch_size, s, d), dtype),
+                 TensorSpec((ubatch_size, s, d), dtype)]
+    elif sub == 2:
+        specs = [TensorSpec((ubatch_size, s, i), dtype),
+                 TensorSpec((ubatch_size, s, d), dtype)]
+    else:
+        specs = [TensorSpec((ubatch_size, s, d), dtype)]
+    if config.model_type == "bert":
+        # padding mask rides every BERT boundary as the LAST tensor
+        # (additive [u, S]; derived from token ids by the first shard)
+        specs.append(TensorSpec((ubatch_size, s), dtype))
+    return specs
 
 
 def output_spec(config: TransformerConfig, ubatch_size: int,

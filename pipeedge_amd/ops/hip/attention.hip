@@ -29,10 +29,14 @@ constexpr int PAD = 8;
 // KVT*row_stride per tile instead of recomputed 64-bit mads (the disassembly
 // showed ~200 VALU address ops per tile; fresh PMC: MfmaUtil 2.9%,
 // VALUBusy 21%, i.e. the kernel was VALU/latency-bound, not memory-bound).
-template <int DHP, int KVT = 64, bool EXACT = false>
+// MASK adds a per-(batch, kpos) additive bias (BERT padding masks; HF's
+// extended [B,1,1,S] mask collapses to [B,S] since it is q-independent).
+// The host pads the mask to a KV-tile multiple so loads need no guards.
+template <int DHP, int KVT = 64, bool EXACT = false, bool MASK = false>
 __global__ __launch_bounds__(256) void qkv_attention_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
-        int B, int S, int H, int dh, float scale) {
+        const float* __restrict__ mask, int B, int S, int H, int dh,
+        float scale) {
     __shared__ __bf16 Qs[QCH][DHP + PAD];
     __shared__ __bf16 Ks[KVT][DHP + PAD];
     __shared__ __bf16 Vs[DHP][KVT + PAD];      // transposed, XOR-swizzled
@@ -184,13 +188,17 @@ __global__ __launch_bounds__(256) void qkv_attention_kernel(
             }
         }
         __builtin_amdgcn_s_setprio(0);
-        // scale + mask invalid kv columns
+        // scale + additive mask + mask invalid kv columns
         #pragma unroll
         for (int nf = 0; nf < KVT / 16; ++nf) {
             const bool valid = s0 + nf * 16 + (lane & 15) < S;
+            float mk = 0.0f;
+            if (MASK)  // host-padded: in-bounds for any tile kpos
+                mk = mask[(long)b * ((S + 127) & ~127)
+                          + s0 + nf * 16 + (lane & 15)];
             #pragma unroll
             for (int r = 0; r < 4; ++r)
-                sfrag[nf][r] = valid ? sfrag[nf][r] * scale : -1e30f;
+                sfrag[nf][r] = valid ? sfrag[nf][r] * scale + mk : -1e30f;
         }
 
         // ---- online softmax update (per C-fragment row r) ----
@@ -318,10 +326,11 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
 // the dh==DHP guards away (the ViT/BERT dh=64 hot case). KROW pads the K
 // LDS row to 128 elements for DHP > 64 so the row stride stays 256 B and
 // the 16-chunk XOR swizzle keeps the A-fragment reads conflict-free.
-template <int KVT, int DHP = 64, bool EXACT = true>
+template <int KVT, int DHP = 64, bool EXACT = true, bool MASK = false>
 __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
-        int B, int S, int H, int dh, float scale) {
+        const float* __restrict__ mask, int B, int S, int H, int dh,
+        float scale) {
     constexpr int QB = 128;                  // q rows per block (4 x 32)
     constexpr int KROW = (DHP <= 64) ? 64 : 128;   // padded K row (elems)
     constexpr int SWZ = (DHP <= 64) ? 7 : 15;      // 16-B chunk XOR mask
@@ -440,6 +449,22 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         for (int mt = 0; mt < KVT / 32; ++mt)
             #pragma unroll
             for (int i = 0; i < 16; ++i) sfrag[mt][i] *= scale;
+        if (MASK) {
+            // lane's kpos(i) = mt*32 + hi*4 + (i&3) + 8*(i>>2): four
+            // contiguous float4 groups per mt; host padding makes the
+            // vector loads in-bounds for every tile position
+            const float* mrow = mask + (long)b * ((S + 127) & ~127) + s0;
+            #pragma unroll
+            for (int mt = 0; mt < KVT / 32; ++mt)
+                #pragma unroll
+                for (int g = 0; g < 4; ++g) {
+                    const f32x4 m4 = *reinterpret_cast<const f32x4*>(
+                        mrow + mt * 32 + hi * 4 + 8 * g);
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        sfrag[mt][g * 4 + j] += m4[j];
+                }
+        }
         if (s0 + KVT > S) {          // ragged tail: mask invalid kpos
             #pragma unroll
             for (int mt = 0; mt < KVT / 32; ++mt)
@@ -727,7 +752,8 @@ __global__ __launch_bounds__(256) void qkv_attention_smallS_kernel(
 }  // namespace
 
 torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
-                                 double scale) {
+                                 double scale,
+                                 c10::optional<torch::Tensor> attn_mask) {
     TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16,
                 "attention: bf16 GPU tensor required");
     TORCH_CHECK(qkv.dim() == 3, "attention: qkv must be [B, S, 3D]");
@@ -742,6 +768,23 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     dim3 grid(B * H, (S + QCH - 1) / QCH);
     const auto* in = reinterpret_cast<const __bf16*>(x.data_ptr());
     auto* out = reinterpret_cast<__bf16*>(ctx.data_ptr());
+
+    // Additive per-(batch, kpos) mask [B, S]: padded to a 128 multiple so
+    // kernel tile loads are unguarded (pad positions add 0 and are then
+    // overridden by the ragged-tail -1e30 masking).
+    const float* mask_ptr = nullptr;
+    torch::Tensor mask_padded;  // keeps the padded copy alive
+    if (attn_mask.has_value()) {
+        auto m = attn_mask.value();
+        TORCH_CHECK(m.is_cuda(), "attention: mask must be a GPU tensor");
+        TORCH_CHECK(m.dim() == 2 && m.size(0) == B && m.size(1) == S,
+                    "attention: mask must be [B, S] (per-kpos additive)");
+        const long sp = (S + 127) & ~127L;
+        mask_padded = torch::zeros({B, sp},
+                                   m.options().dtype(torch::kFloat32));
+        mask_padded.narrow(1, 0, S).copy_(m.to(torch::kFloat32));
+        mask_ptr = mask_padded.data_ptr<float>();
+    }
     static const int kvt = []() {
         const char* e = std::getenv("PIPEEDGE_ATTN_KVT");
         // KVT=128 is ~1us faster SOLO, but KVT=64 (28 KiB LDS after the
@@ -756,6 +799,34 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         const char* e = std::getenv("PIPEEDGE_ATTN_V2");
         return e ? std::atoi(e) : 1;
     }();
+// launch helpers: pick the MASK template arm from mask_ptr
+#define PE_LAUNCH_V2(KVT_, DHP_, EXACT_)                                     \
+    do {                                                                     \
+        if (mask_ptr)                                                        \
+            hipLaunchKernelGGL(                                              \
+                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, true>), grid2,  \
+                dim3(256), 0, pe_current_stream(), in, out, mask_ptr, B, S,  \
+                H, dh, (float)scale);                                        \
+        else                                                                 \
+            hipLaunchKernelGGL(                                              \
+                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, false>), grid2, \
+                dim3(256), 0, pe_current_stream(), in, out, nullptr, B, S,   \
+                H, dh, (float)scale);                                        \
+    } while (0)
+#define PE_LAUNCH_V1(DHP_, KVT_, EXACT_)                                     \
+    do {                                                                     \
+        if (mask_ptr)                                                        \
+            hipLaunchKernelGGL(                                              \
+                (qkv_attention_kernel<DHP_, KVT_, EXACT_, true>), grid,      \
+                dim3(256), 0, pe_current_stream(), in, out, mask_ptr, B, S,  \
+                H, dh, (float)scale);                                        \
+        else                                                                 \
+            hipLaunchKernelGGL(                                              \
+                (qkv_attention_kernel<DHP_, KVT_, EXACT_, false>), grid,     \
+                dim3(256), 0, pe_current_stream(), in, out, nullptr, B, S,   \
+                H, dh, (float)scale);                                        \
+    } while (0)
+
     if (dh == 64 && v2) {
         dim3 grid2(B * H, (S + 127) / 128);
         // KV tile length: 128 halves the barrier count and wins at ViT-class
@@ -765,13 +836,9 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         // forces 64
         const bool kvt128 = v2 == 2 || (v2 == 1 && S > 64 && S <= 256);
         if (kvt128)
-            hipLaunchKernelGGL((qkv_attention_v2_kernel<128, 64, true>),
-                               grid2, dim3(256), 0, pe_current_stream(), in,
-                               out, B, S, H, dh, (float)scale);
+            PE_LAUNCH_V2(128, 64, true);
         else
-            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 64, true>),
-                               grid2, dim3(256), 0, pe_current_stream(), in,
-                               out, B, S, H, dh, (float)scale);
+            PE_LAUNCH_V2(64, 64, true);
         HIP_CHECK_LAST();
         return ctx;
     }
@@ -779,13 +846,9 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         // ViT-Huge class (dh=80): zero-padded to DHP=96
         dim3 grid2(B * H, (S + 127) / 128);
         if (dh == 96)
-            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 96, true>),
-                               grid2, dim3(256), 0, pe_current_stream(), in,
-                               out, B, S, H, dh, (float)scale);
+            PE_LAUNCH_V2(64, 96, true);
         else
-            hipLaunchKernelGGL((qkv_attention_v2_kernel<64, 96, false>),
-                               grid2, dim3(256), 0, pe_current_stream(), in,
-                               out, B, S, H, dh, (float)scale);
+            PE_LAUNCH_V2(64, 96, false);
         HIP_CHECK_LAST();
         return ctx;
     }
@@ -797,35 +860,27 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         const char* e = std::getenv("PIPEEDGE_ATTN_SMALLS");
         return e != nullptr && std::atoi(e) != 0;
     }();
-    if (dh <= 64 && S <= 256 && small_s) {
+    if (dh <= 64 && S <= 256 && small_s && !mask_ptr) {
         hipLaunchKernelGGL((qkv_attention_smallS_kernel<64, 256>), grid,
                            dim3(256), 0, pe_current_stream(), in, out,
                            B, S, H, dh, (float)scale);
     } else if (dh == 64) {
         if (kvt >= 128 && S >= 128) {
-            hipLaunchKernelGGL((qkv_attention_kernel<64, 128, true>), grid,
-                               dim3(256), 0, pe_current_stream(), in, out,
-                               B, S, H, dh, (float)scale);
+            PE_LAUNCH_V1(64, 128, true);
         } else {
-            hipLaunchKernelGGL((qkv_attention_kernel<64, 64, true>), grid,
-                               dim3(256), 0, pe_current_stream(), in, out,
-                               B, S, H, dh, (float)scale);
+            PE_LAUNCH_V1(64, 64, true);
         }
     } else if (dh < 64) {
-        hipLaunchKernelGGL((qkv_attention_kernel<64, 64>), grid,
-                           dim3(256), 0, pe_current_stream(), in, out,
-                           B, S, H, dh, (float)scale);
+        PE_LAUNCH_V1(64, 64, false);
     } else if (dh <= 96) {
-        hipLaunchKernelGGL((qkv_attention_kernel<96>), grid, dim3(256), 0,
-                           pe_current_stream(), in, out, B, S, H, dh,
-                           (float)scale);
+        PE_LAUNCH_V1(96, 64, false);
     } else if (dh <= 128) {
-        hipLaunchKernelGGL((qkv_attention_kernel<128>), grid, dim3(256), 0,
-                           pe_current_stream(), in, out, B, S, H, dh,
-                           (float)scale);
+        PE_LAUNCH_V1(128, 64, false);
     } else {
         TORCH_CHECK(false, "attention: head dim > 128 unsupported");
     }
     HIP_CHECK_LAST();
     return ctx;
 }
+#undef PE_LAUNCH_V2
+#undef PE_LAUNCH_V1

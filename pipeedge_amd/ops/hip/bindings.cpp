@@ -20,7 +20,8 @@ torch::Tensor add_layernorm_bf16(torch::Tensor x, torch::Tensor residual,
                                  torch::Tensor weight, torch::Tensor bias,
                                  double eps);
 torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
-                                 double scale);
+                                 double scale,
+                                 c10::optional<torch::Tensor> attn_mask);
 torch::Tensor patchify_bf16(torch::Tensor pixels, int64_t patch_size);
 torch::Tensor embed_finish_bf16(torch::Tensor x, torch::Tensor prefix,
                                 torch::Tensor pos);
@@ -59,9 +60,15 @@ std::tuple<torch::Tensor, torch::Tensor> linear_op(
 torch::Tensor qkv_attention_op(torch::Tensor qkv, int64_t num_heads,
                                double scale,
                                c10::optional<torch::Tensor> attn_mask) {
-    TORCH_CHECK(!attn_mask.has_value(),
-                "HIP attention: attn_mask not supported yet");
-    return qkv_attention_bf16(qkv, num_heads, scale);
+    // mask: [B, S] additive bias per kv position (BERT padding); broadcast
+    // shapes [B,1,1,S] / [B,1,S] collapse to [B,S]
+    if (attn_mask.has_value()) {
+        auto m = attn_mask.value();
+        while (m.dim() > 2 && m.size(1) == 1)
+            m = m.squeeze(1);
+        attn_mask = m.contiguous();
+    }
+    return qkv_attention_bf16(qkv, num_heads, scale, attn_mask);
 }
 
 }  // namespace

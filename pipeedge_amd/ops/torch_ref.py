@@ -78,6 +78,10 @@ def qkv_attention(qkv: Tensor, num_heads: int, scale: Optional[float] = None,
     def _heads(t):
         return t.reshape(b, s, num_heads, dh).permute(0, 2, 1, 3)
 
+    if attn_mask is not None and attn_mask.dim() == 2:
+        # [B, S] per-kpos additive mask (the HIP kernel's contract):
+        # broadcast over heads and query rows
+        attn_mask = attn_mask.reshape(b, 1, 1, s)
     ctx = attention_core(_heads(q), _heads(k), _heads(v), scale, attn_mask)
     return ctx.permute(0, 2, 1, 3).reshape(b, s, d)
 

@@ -61,7 +61,10 @@ def load_dataset_synthetic(model_config, batch_size: int,
                                (batch_size,), generator=gen)
     else:
         s = seq_len or model_config.max_position_embeddings
-        data = torch.randint(0, model_config.vocab_size, (batch_size, s),
+        # ids start past the special-token range so no synthetic position
+        # collides with pad_token_id (full-length sequences, no padding)
+        lo = min(1000, model_config.vocab_size - 1)
+        data = torch.randint(lo, model_config.vocab_size, (batch_size, s),
                              generator=gen)
         labels = torch.randint(0, max(model_config.num_labels, 2),
                                (batch_size,), generator=gen)

@@ -139,3 +139,50 @@ def test_vit_huge_head_dim_80_shard():
     shard = vit.ViTShardForImageClassification(cfg, _shard_cfg(1, 4, 128), None)
     out = shard(torch.randn(1, 3, 224, 224))
     assert out.shape == (1, 257, 1280)
+
+
+def test_bert_padding_mask_matches_unpadded():
+    """Padded-batch outputs must equal per-sequence unpadded outputs: the
+    first shard derives the additive mask from pad ids and attention ignores
+    padding (the reference's shards run mask-free over padding and get this
+    wrong - reference bert.py:41-52)."""
+    cfg = get_config("textattack/bert-base-uncased-CoLA")
+    total = 48
+    full = bert.BertShardForSequenceClassification(
+        cfg, _shard_cfg(1, total, total), None)
+    gen = torch.Generator().manual_seed(7)
+    s_full, s_short = 64, 40
+    ids = torch.randint(1000, cfg.vocab_size, (2, s_full), generator=gen)
+    ids[1, s_short:] = cfg.pad_token_id  # sequence 1 padded
+    y_padded = full(ids)
+    # unpadded reference for sequence 1: same tokens, no pad positions
+    y_short = full(ids[1:2, :s_short])
+    assert torch.allclose(y_padded[1], y_short[0], atol=1e-4, rtol=1e-4), \
+        (y_padded[1] - y_short[0]).abs().max()
+    # and sequence 0 (no padding) unchanged vs its own unpadded run
+    # (allclose: batch-1 vs batch-2 matmul reduction order differs on CPU)
+    y0 = full(ids[0:1])
+    assert torch.allclose(y_padded[0], y0[0], atol=1e-6, rtol=1e-6)
+
+
+@pytest.mark.parametrize("partition", [
+    [(1, 24), (25, 48)],
+    [(1, 2), (3, 13), (14, 14), (15, 48)],  # mid-block boundaries
+])
+def test_bert_masked_stitched_equals_full(partition):
+    """The mask must survive every stage boundary type (1-, 2-tensor
+    payloads + trailing mask)."""
+    cfg = get_config("textattack/bert-base-uncased-CoLA")
+    full = bert.BertShardForSequenceClassification(
+        cfg, _shard_cfg(1, 48, 48), None)
+    gen = torch.Generator().manual_seed(8)
+    ids = torch.randint(1000, cfg.vocab_size, (2, 48), generator=gen)
+    ids[0, 30:] = cfg.pad_token_id
+    ids[1, 17:] = cfg.pad_token_id
+    y_full = full(ids)
+    data = ids
+    for a, b in partition:
+        shard = bert.BertShardForSequenceClassification(
+            cfg, _shard_cfg(a, b, 48), None)
+        data = shard(data)
+    assert torch.equal(data, y_full)
