@@ -392,6 +392,122 @@ __global__ __launch_bounds__(256) void gemm_bf16_glds3_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// 8-wave big-tile pipeline (512 threads, wave grid 2(M) x 4(N)): the
+// large-shape kernel (ViT-H fc1 / 4k^3 class) where the 256-thread 64x64
+// tiles are ~3x off the library (profiles/blaslt_probe.txt). Structure per
+// the guide's measured tiers (cdna_hip_programming.md section 5): BK=64
+// glds staging, raw s_barrier + counted vmcnt (NBUF=3: one tile of LDS-DMA
+// in flight across each barrier), XOR source swizzle; 512 threads put TWO
+// waves on every SIMD so MFMA issue and LDS/DMA segments of partner waves
+// overlap (MI355X_MICROARCH "Two waves per SIMD").
+//   <4,4,3>: 128x256 tile, 144 KiB LDS, 1 block/CU
+//   <8,4,2>: 256x256 tile, 128 KiB LDS, 1 block/CU (classic 2-buffer)
+// ---------------------------------------------------------------------------
+template <int MI, int NI, int NBUF, bool XCD = true>
+__global__ __launch_bounds__(512) void gemm_bf16_glds8_kernel(
+        const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+        __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
+    constexpr int TBM = 2 * 16 * MI;     // 2 wave-rows
+    constexpr int TBN = 4 * 16 * NI;     // 4 wave-cols
+    constexpr int KB = BK;
+    __shared__ __bf16 smem[NBUF][TBM + TBN][KB];
+
+    int bx = blockIdx.x, by = blockIdx.y;
+    if (XCD) {
+        const int nb = gridDim.x * gridDim.y;
+        const int id = xcd_swizzle(by * gridDim.x + bx, nb);
+        bx = id % gridDim.x;
+        by = id / gridDim.x;
+    }
+    const int bm = bx * TBM;
+    const int bn = by * TBN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;            // 0..7
+    const int wm = (wid >> 2) * (16 * MI);
+    const int wn = (wid & 3) * (16 * NI);
+
+    const int srow = lane >> 3;          // row within an 8-row piece
+    const int cdst = lane & 7;           // 16-B column chunk
+
+    // pieces of 8 rows x 64 elems; (TBM+TBN)/8 pieces split across 8 waves
+    constexpr int PIECES_A = TBM / 8 / 8;   // per wave
+    constexpr int PIECES_B = TBN / 8 / 8;
+    constexpr int GLDS_PER_TILE = PIECES_A + PIECES_B;
+    auto stage = [&](int buf, int k0) {
+        #pragma unroll
+        for (int i = 0; i < PIECES_A; ++i) {
+            const int p = wid * PIECES_A + i;
+            const int trow = p * 8 + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long arow = min(bm + trow, M - 1);
+            glds_b128(A + arow * K + k0 + csrc * 8, &smem[buf][p * 8][0]);
+        }
+        #pragma unroll
+        for (int i = 0; i < PIECES_B; ++i) {
+            const int p = wid * PIECES_B + i;
+            const int trow = p * 8 + srow;
+            const int csrc = cdst ^ (trow & 7);
+            const long brow = min(bn + trow, N - 1);
+            glds_b128(B + brow * K + k0 + csrc * 8,
+                      &smem[buf][TBM + p * 8][0]);
+        }
+    };
+
+    f32x4 acc[MI][NI] = {};
+    const int n_tiles = K / KB;
+    #pragma unroll
+    for (int p = 0; p < NBUF - 1; ++p)
+        if (p < n_tiles) stage(p, p * KB);
+
+    // Same raw-barrier counted-vmcnt loop as the 256-thread glds3 kernel
+    // (a stage-before-wait restructure with an extra barrier measured
+    // SLOWER - sq4k 132->143us - because the post-barrier stage already
+    // flies under compute; keep the proven structure).
+    for (int t = 0; t < n_tiles; ++t) {
+        const int inflight = min(NBUF - 2, n_tiles - 1 - t);
+        switch (inflight) {
+        case 0:
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            break;
+        case 1:
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GLDS_PER_TILE)
+                         : "memory");
+            break;
+        default:
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GLDS_PER_TILE)
+                         : "memory");
+            break;
+        }
+        __builtin_amdgcn_s_barrier();
+        if (t + NBUF - 1 < n_tiles)
+            stage((t + NBUF - 1) % NBUF, (t + NBUF - 1) * KB);
+        const int buf = t % NBUF;
+        #pragma unroll
+        for (int kk = 0; kk < KB; kk += 32) {
+            const int kgrp = kk + (lane >> 4) * 8;
+            bf16x8 af[MI], bfr[NI];
+            #pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+                af[mi] = lds_read_swz<KB>(&smem[buf][0][0],
+                                          wm + mi * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+                bfr[ni] = lds_read_swz<KB>(&smem[buf][TBM][0],
+                                           wn + ni * 16 + (lane & 15), kgrp);
+            #pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+                #pragma unroll
+                for (int ni = 0; ni < NI; ++ni)
+                    acc[mi][ni] = mfma_16x16x32_bf16(af[mi], bfr[ni],
+                                                     acc[mi][ni]);
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    gemm_epilogue<MI, NI, 0>(acc, ep, bm, bn, wm, wn, lane, 0, nullptr);
+}
+
+// ---------------------------------------------------------------------------
 // generic path: any K (padded LDS, plain vectorized staging)
 // ---------------------------------------------------------------------------
 template <int EPM = 0>
@@ -533,15 +649,17 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
             t = env_t;
         }
         if (t == 0) {
-            // from the on-HW sweeps (tools/gemm_sweep.py, hotpath_sweep.py,
-            // glds3_screen.py, bigshape_sweep.py): XCD-swizzled 64x64
-            // 2-buffer default; the 3-buffer pipeline (t7) for the
-            // mid-size long-K / wide-N shapes (ViT-B/L fc1/fc2 class);
-            // 128-row tiles once M >= 2048 (ViT-H class: t2 beats t7 by
-            // 15% on fc1, t5 beats t7 by 18% on fc2); 128x128 for the
-            // large near-square regime (4k^3: t3 is 2x t7)
-            if (M >= 4096 && N >= 4096 && K >= 2048) t = 3;
-            else if (M >= 2048) t = (N >= 3072) ? 2 : 5;
+            // From the on-HW sweeps (round 2: tools/bigtile_sweep.py medians,
+            // profiles/bigtile_sweep_r02.txt): the 512-thread 8-wave 256x256
+            // 2-buffer kernel (t13) for the big near-square and ViT-H fc1
+            // regimes (sq4k 127us vs r1-auto 162; vitH-fc1 55 vs 62; the r1
+            // choice t2 also showed multi-hundred-us pathologies there);
+            // 256x128 3-buffer (t14) for the ViT-L fc1 class; 3-buffer
+            // 64x64 (t7) for mid-size long-K; XCD-swizzled 64x64 (t5)
+            // otherwise.
+            if (M >= 3072 && N >= 3072 && K >= 2048) t = 13;
+            else if (M >= 2048) t = (N >= 3072) ? 13 : 5;
+            else if (M >= 1024 && N >= 4096 && K < 2048) t = 14;
             else if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
             else t = 5;
         }
@@ -593,7 +711,26 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 (gemm_bf16_glds3_kernel<MI_, NI_, true, 0, NBUF_>), GRID_,   \
                 dim3(256), 0, pe_current_stream(), ap, bp, cp, M, N, K, ep); \
     } while (0)
-        if (t == 7) {
+        if ((t == 12 || t == 13 || t == 14) && epm == 0) {
+            // 8-wave big-tile kernels (epilogue modes 1/2 unsupported here;
+            // the auto heuristic only picks these for plain big shapes)
+            if (t == 12) {
+                dim3 grid((M + 127) / 128, (N + 255) / 256);
+                hipLaunchKernelGGL((gemm_bf16_glds8_kernel<4, 4, 3>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
+            } else if (t == 13) {
+                dim3 grid((M + 255) / 256, (N + 255) / 256);
+                hipLaunchKernelGGL((gemm_bf16_glds8_kernel<8, 4, 2>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
+            } else {
+                dim3 grid((M + 255) / 256, (N + 127) / 128);
+                hipLaunchKernelGGL((gemm_bf16_glds8_kernel<8, 2, 3>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
+            }
+        } else if (t == 7 || t == 12 || t == 13 || t == 14) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             PE_LAUNCH_GLDS3(2, 2, 3, grid);
         } else if (t == 10) {
