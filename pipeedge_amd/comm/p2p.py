@@ -102,7 +102,13 @@ class DistP2pContext(DistContext):
                  backend: Optional[str] = None):
         super().__init__(world_size, rank)
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            # PIPEEDGE_P2P_BACKEND=gloo forces a CPU-staged wire with GPU
+            # compute - lets N pipeline stages share one GPU for validation
+            # (RCCL forbids two ranks per device)
+            import os
+            backend = os.environ.get(
+                "PIPEEDGE_P2P_BACKEND",
+                "nccl" if torch.cuda.is_available() else "gloo")
         self._backend = backend
         self._cmd_cb = cmd_cb
         self._ctrl_group = None

@@ -693,8 +693,19 @@ def run_pipeline_rpc(*args, **kwargs) -> None:
 
 
 def init_env(device: Optional[str], net_addr: str, net_port: int,
-             net_ifname: str) -> None:
-    """Initialize the runtime environment (reference runtime.py:581-602)."""
+             net_ifname: str, rank: int = 0) -> None:
+    """Initialize the runtime environment (reference runtime.py:581-602).
+
+    A bare ``cuda`` device (or default) on a multi-GPU host maps this rank
+    to ``cuda:<LOCAL_RANK or rank % device_count>`` - one GPU per rank, the
+    RCCL requirement (single-node multi-rank; the reference ran one process
+    per HOST so never needed this).
+    """
+    if device in (None, "cuda") and torch.cuda.is_available() \
+            and torch.cuda.device_count() > 1:
+        local = int(os.environ.get("LOCAL_RANK",
+                                   rank % torch.cuda.device_count()))
+        device = f"cuda:{local}"
     devices.initialize(device)
     os.environ.setdefault("MASTER_ADDR", net_addr)
     os.environ.setdefault("MASTER_PORT", str(net_port))
@@ -785,7 +796,8 @@ def main() -> None:
     hosts = None if args.hosts is None else args.hosts.split(",")
 
     tik = time.time()
-    init_env(args.device, args.addr, args.port, args.socket_ifname)
+    init_env(args.device, args.addr, args.port, args.socket_ifname,
+             rank=args.rank)
     logger.info("Device: %s", devices.DEVICE)
     if args.comm == "p2p":
         run_pipeline_p2p(args.worldsize, args.rank, args.model_name,
