@@ -334,9 +334,14 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     constexpr int QB = 128;                  // q rows per block (4 x 32)
     constexpr int KROW = (DHP <= 64) ? 64 : 128;   // padded K row (elems)
     constexpr int SWZ = (DHP <= 64) ? 7 : 15;      // 16-B chunk XOR mask
-    constexpr int NT = DHP / 32;             // 32-wide d tiles
+    // DHP need only be a 16-multiple: scores/Q/K/staging run at the NATIVE
+    // head extent (DHP=80 for ViT-Huge - round 1 zero-padded everything to
+    // 96, wasting 20% of QK^T MFMAs and staging); only the PV d-tiles stay
+    // 32-wide (VW pads the V image; pad-block outputs are never stored).
+    constexpr int NT = (DHP + 31) / 32;      // 32-wide d tiles (PV/output)
+    constexpr int VW = NT * 32;              // padded V image width
     __shared__ __bf16 Ks[KVT][KROW];         // row-major, XOR-swizzled cols
-    __shared__ __bf16 Vs[KVT * DHP];         // [s/4][d/16][4][16] blocked
+    __shared__ __bf16 Vs[KVT * VW];          // [s/4][d/16][4][16] blocked
 
     const int bh = blockIdx.x;
     const int b = bh / H;
@@ -369,16 +374,21 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     }
 
     // ---- K/V staging: per-thread running pointers; row/col from the
-    // bijective piece index (DHP/8 need not divide 256) ----
-    constexpr int PIECES = KVT * DHP / 8 / 256;
+    // bijective piece index (DHP/8 need not divide 256; the piece total
+    // need not be a 256-multiple either - DHP=80 at KVT=64 has 640 pieces,
+    // the top half of the last round is idle) ----
+    constexpr int TOTAL_PIECES = KVT * DHP / 8;
+    constexpr int PIECES = (TOTAL_PIECES + 255) / 256;
+    constexpr bool RAGGED = (TOTAL_PIECES % 256) != 0;
     int lr[PIECES], lc[PIECES];
     const __bf16* k_pc[PIECES];
     const __bf16* v_pc[PIECES];
     #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
         const int piece = tid + i * 256;
-        lr[i] = piece / (DHP / 8);
-        lc[i] = (piece % (DHP / 8)) * 8;
+        const int p = RAGGED ? min(piece, TOTAL_PIECES - 1) : piece;
+        lr[i] = p / (DHP / 8);
+        lc[i] = (p % (DHP / 8)) * 8;
         k_pc[i] = k_base + (long)lr[i] * row_stride + lc[i];
         v_pc[i] = v_base + (long)lr[i] * row_stride + lc[i];
     }
@@ -387,6 +397,7 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         const int s0_ = t * KVT;
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
+            if (RAGGED && tid + i * 256 >= TOTAL_PIECES) continue;
             const int gs = s0_ + lr[i];
             k_reg[i] = bf16x8{};
             v_reg[i] = bf16x8{};
@@ -401,13 +412,14 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     auto write_tile = [&]() {
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
+            if (RAGGED && tid + i * 256 >= TOTAL_PIECES) continue;
             const int r = lr[i];
             // K: swizzle the 16-B column chunk by row&SWZ (reads match)
             const int kchunk = (lc[i] / 8) ^ (r & SWZ);
             *reinterpret_cast<bf16x8*>(&Ks[r][kchunk * 8]) = k_reg[i];
             // V: blocked tile (s/4, d/16), row s%4, col d%16 - the thread's
             // 8 d-elements are contiguous within one 16-wide tile row
-            const int vtile = (r / 4) * (DHP / 16) + lc[i] / 16;
+            const int vtile = (r / 4) * (VW / 16) + lc[i] / 16;
             *reinterpret_cast<bf16x8*>(
                 &Vs[vtile * 64 + (r % 4) * 16 + (lc[i] % 16)]) = v_reg[i];
         }
@@ -530,11 +542,11 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
                 auto* t1 = reinterpret_cast<
                     __attribute__((address_space(3))) v4bf*>(
                     reinterpret_cast<uintptr_t>(&Vs[0]) +
-                    (sb1 * (DHP / 16) + db) * 128 + (lane & 15) * 8);
+                    (sb1 * (VW / 16) + db) * 128 + (lane & 15) * 8);
                 auto* t2 = reinterpret_cast<
                     __attribute__((address_space(3))) v4bf*>(
                     reinterpret_cast<uintptr_t>(&Vs[0]) +
-                    ((sb1 + 1) * (DHP / 16) + db) * 128 + (lane & 15) * 8);
+                    ((sb1 + 1) * (VW / 16) + db) * 128 + (lane & 15) * 8);
                 const v4bf r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t1);
                 const v4bf r2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(t2);
                 bf16x8 av;
@@ -842,8 +854,17 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         HIP_CHECK_LAST();
         return ctx;
     }
+    if (dh == 80 && v2) {
+        // ViT-Huge: NATIVE DHP=80 scores/Q/K/staging (round 1 zero-padded
+        // to 96: 20% wasted QK^T MFMAs + staging); KVT=64 keeps the 28 KiB
+        // LDS footprint (5 blocks/CU)
+        dim3 grid2(B * H, (S + 127) / 128);
+        PE_LAUNCH_V2(64, 80, false);
+        HIP_CHECK_LAST();
+        return ctx;
+    }
     if (dh > 64 && dh <= 96 && dh % 8 == 0 && v2) {
-        // ViT-Huge class (dh=80): zero-padded to DHP=96
+        // other padded head dims: zero-padded to DHP=96
         dim3 grid2(B * H, (S + 127) / 128);
         if (dh == 96)
             PE_LAUNCH_V2(64, 96, true);
