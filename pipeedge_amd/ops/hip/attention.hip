@@ -841,12 +841,13 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
 
     if (dh == 64 && v2) {
         dim3 grid2(B * H, (S + 127) / 128);
-        // KV tile length: 128 halves the barrier count and wins at ViT-class
-        // sequences (9.1 vs 9.7us solo; e2e 10,500), but its doubled score
-        // register state loses at long sequences (BERT s384: 5,290 vs
-        // 5,525) - auto-pick by S; PIPEEDGE_ATTN_V2=2 forces 128, =3
-        // forces 64
-        const bool kvt128 = v2 == 2 || (v2 == 1 && S > 64 && S <= 256);
+        // KV tile length: 128 halves the barrier count and wins SOLO at
+        // ViT-class sequences (9.1 vs 9.7us), but 64's 16 KiB LDS doubles
+        // co-residency (8 vs 5 blocks/CU) and wins the 2-stream e2e bench
+        // in every interleaved A/B round (r2: 10.24-10.38k vs 9.95-10.13k
+        // img/s, profiles/kvt_ab_r02.txt) - default 64;
+        // PIPEEDGE_ATTN_V2=2 forces 128, =3 forces 64
+        const bool kvt128 = v2 == 2;
         if (kvt128)
             PE_LAUNCH_V2(128, 64, true);
         else
