@@ -114,3 +114,23 @@ def load_dataset_imagenet(root: str, split: str, batch_size: int,
             break
     return RolloverTensorDataset(batch_size, torch.stack(data),
                                  torch.tensor(labels))
+
+
+def load_dataset_npz(path: str, batch_size: int) -> Dataset:
+    """Tokenized-input dataset from an ``.npz`` file (offline analogue of
+    the reference's tools/bert_save_input.py flow, which stores raw CoLA
+    sentences and tokenizes at load - tokenizer vocabularies need network
+    access, so this environment stores token ids directly).
+
+    Keys: ``ids`` int64 [N, S] (padded with the model's pad id) and
+    optional ``label`` [N].
+    """
+    import numpy as np
+    data = np.load(path, allow_pickle=True)
+    key = "ids" if "ids" in data else "input"
+    ids = torch.as_tensor(np.asarray(data[key], dtype="int64"))
+    if "label" in data:
+        labels = torch.as_tensor(np.asarray(data["label"], dtype="int64"))
+    else:
+        labels = torch.zeros(ids.shape[0], dtype=torch.int64)
+    return RolloverTensorDataset(batch_size, ids, labels)

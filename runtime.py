@@ -489,6 +489,9 @@ def load_dataset(dataset_cfg: dict, model_name: str, batch_size: int,
         dataset = load_dataset_glue(model_name, "cola",
                                     dataset_cfg.get("split", "train"),
                                     batch_size, indices=indices)
+    elif name == "npz":
+        from pipeedge_amd.utils.data import load_dataset_npz
+        dataset = load_dataset_npz(dataset_cfg["root"], batch_size)
     elif name == "ImageNet":
         from pipeedge_amd.utils.data import load_dataset_imagenet
         dataset = load_dataset_imagenet(dataset_cfg["root"],
@@ -745,7 +748,7 @@ def main() -> None:
                         help="number of batches to stream (throughput runs)")
     dset = parser.add_argument_group("Dataset arguments")
     dset.add_argument("--dataset-name", type=str,
-                      choices=["CoLA", "ImageNet", "synthetic"],
+                      choices=["CoLA", "ImageNet", "npz", "synthetic"],
                       help="dataset to use (default: synthetic random)")
     dset.add_argument("--dataset-root", type=str,
                       help="dataset root directory")

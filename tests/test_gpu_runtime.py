@@ -134,3 +134,28 @@ def test_bench_two_gpu_contract():
             if l.startswith("{")][-1]
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["value"] > 0
+
+
+def test_runtime_4stage_bert_cola_variable_length_shared_gpu(tmp_path):
+    """BASELINE config #4 workload on hardware: BERT-CoLA classifier,
+    4 pipeline stages (one GPU, gloo-staged wire), variable-length padded
+    batch from a tokenized npz - attention masks padding end to end."""
+    import numpy as np
+    rng = np.random.default_rng(9)
+    n, smax = 32, 128
+    ids = np.zeros((n, smax), dtype=np.int64)
+    for i in range(n):
+        ln = int(rng.integers(16, smax + 1))
+        ids[i, :ln] = rng.integers(1000, 30000, size=ln)
+    np.savez(tmp_path / "cola.npz", ids=ids,
+             label=rng.integers(0, 2, size=n))
+    env = {"PIPEEDGE_P2P_BACKEND": "gloo", "PIPEEDGE_UBATCH_STREAMS": "0"}
+    outs = _run_multi(
+        4, ["-d", "cuda", "-m", "textattack/bert-base-uncased-CoLA",
+            "-b", "32", "-u", "8", "-pt", "1,12,13,24,25,36,37,48",
+            "--dataset-name", "npz",
+            "--dataset-root", str(tmp_path / "cola.npz")],
+        env_extra=env, port=29791)
+    for r in range(4):
+        assert outs[r][0] == 0, (r, outs[r][2][-2000:])
+    assert "throughput is" in outs[0][1]

@@ -185,3 +185,27 @@ def test_runtime_reversed_rank_order():
     for r in range(2):
         assert outs[r][0] == 0, (r, outs[r][2])
     assert "throughput is" in outs[1][1]
+
+
+def test_runtime_bert_variable_length_npz(tmp_path):
+    """BASELINE config #4 workload shape: BERT-CoLA classifier over a
+    VARIABLE-LENGTH padded batch (tokenized npz from tools/bert_save_input
+    --tokenized); padding masked by the first shard (see bert.py)."""
+    import numpy as np
+    rng = np.random.default_rng(3)
+    n, smax = 16, 64
+    ids = np.zeros((n, smax), dtype=np.int64)
+    for i in range(n):
+        ln = int(rng.integers(8, smax + 1))
+        ids[i, :ln] = rng.integers(1000, 30000, size=ln)
+    np.savez(tmp_path / "cola.npz", ids=ids,
+             label=rng.integers(0, 2, size=n))
+    outs = _run_two(29643, ["-b", "8", "-u", "2",
+                            "-m", "textattack/bert-base-uncased-CoLA",
+                            "-pt", "1,24,25,48",
+                            "--dataset-name", "npz",
+                            "--dataset-root", str(tmp_path / "cola.npz")],
+                    cwd=str(tmp_path))
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]

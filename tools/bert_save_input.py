@@ -17,7 +17,23 @@ def main():
     parser.add_argument("-o", "--output", default="bert_input.npz")
     parser.add_argument("-n", "--num-inputs", type=int, default=512)
     parser.add_argument("--use-glue", action="store_true")
+    parser.add_argument("--tokenized", action="store_true",
+                        help="write token IDS directly (variable-length "
+                             "sequences padded with id 0) - usable offline "
+                             "by `runtime.py --dataset-name npz`")
+    parser.add_argument("--max-len", type=int, default=128)
     args = parser.parse_args()
+    if args.tokenized:
+        rng = np.random.default_rng(0)
+        n, smax = args.num_inputs, args.max_len
+        ids = np.zeros((n, smax), dtype=np.int64)  # 0 = [PAD]
+        for i in range(n):
+            ln = int(rng.integers(8, smax + 1))
+            ids[i, :ln] = rng.integers(1000, 30000, size=ln)
+        label = rng.integers(0, 2, size=n)
+        np.savez(args.output, ids=ids, label=label)
+        print(f"wrote {args.output} (tokenized, variable-length)")
+        return
     if args.use_glue:
         import datasets
         ds_train = datasets.load_dataset("glue", name="cola", split="train")
