@@ -892,7 +892,7 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 return s ? std::atoi(s) : 0;
             }();
             if (M >= 3072 && N >= 3072 && K >= 2048) t = 13;
-            else if (M >= 2048) t = (N >= 3072) ? 13 : 5;
+            else if (M >= 2048) t = (N >= 3072) ? 15 : 5;
             else if (M >= 1024 && N >= 4096 && K < 2048) t = 14;
             else if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
             else if (midt != 0 && M >= 256) t = midt;
@@ -984,7 +984,15 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                                    dim3(512), 0, pe_current_stream(), ap, bp,
                                    cp, M, N, K, ep);
             }
-        } else if (t == 7 || t == 12 || t == 13 || t == 14 || t == 20) {
+        } else if (t == 15 && epm == 0) {
+            // 256x192 2-buffer (112 KiB LDS): ViT-H fc1 grid = 9x27 = 243
+            // blocks ~ one full chip wave (256x256 gives only 180)
+            dim3 grid((M + 255) / 256, (N + 191) / 192);
+            hipLaunchKernelGGL((gemm_bf16_glds8_kernel<8, 3, 2>), grid,
+                               dim3(512), 0, pe_current_stream(), ap, bp,
+                               cp, M, N, K, ep);
+        } else if (t == 7 || t == 12 || t == 13 || t == 14 || t == 15 ||
+                   t == 20) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             PE_LAUNCH_GLDS3(2, 2, 3, grid);
         } else if (t == 10) {

@@ -29,14 +29,14 @@ def bench(fn, iters=40):
         ts.append((time.perf_counter() - t0) * 1e6)
     return sorted(ts)[iters // 2]
 
-hdr = f"{'shape':9s} {'M':>5} {'N':>5} {'K':>5} |{'auto':>8}{'t7':>8}{'t12':>8}{'t13':>8}{'t14':>8}{'lib':>8} us | best TF  (best/lib)"
+hdr = f"{'shape':9s} {'M':>5} {'N':>5} {'K':>5} |{'auto':>8}{'t7':>8}{'t12':>8}{'t13':>8}{'t14':>8}{'t15':>8}{'lib':>8} us | best TF  (best/lib)"
 print(hdr)
 for name, m, n, k in shapes:
     a = (torch.randn(m, k) * 0.3).bfloat16().cuda()
     b = (torch.randn(n, k) * 0.1).bfloat16().cuda()
     ref = torch.matmul(a.float(), b.float().t())
     res = {}
-    for label, tl in (("auto", 0), ("t7", 7), ("t12", 12), ("t13", 13), ("t14", 14)):
+    for label, tl in (("auto", 0), ("t7", 7), ("t12", 12), ("t13", 13), ("t14", 14), ("t15", 15)):
         out, _ = hip.linear(a, b, tile=tl)
         err = (out.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1)
         if err > 2e-2:
@@ -48,7 +48,7 @@ for name, m, n, k in shapes:
     fl = 2 * m * n * k
     best = min(v for kk, v in res.items() if kk != "lib" and v == v)
     print(f"{name:9s} {m:>5} {n:>5} {k:>5} |{res['auto']:8.1f}{res['t7']:8.1f}{res['t12']:8.1f}"
-          f"{res['t13']:8.1f}{res['t14']:8.1f}{res['lib']:8.1f} us | {fl/best/1e6:7.0f} TF  "
+          f"{res['t13']:8.1f}{res['t14']:8.1f}{res['t15']:8.1f}{res['lib']:8.1f} us | {fl/best/1e6:7.0f} TF  "
           f"({res['lib']/best:.2f}x)")
     del a, b, ref
     torch.cuda.empty_cache()

@@ -18,15 +18,17 @@ shapes = [
 ]
 
 def bench(fn, iters=60):
+    """Per-iteration MEDIAN (rare multi-ms system stalls poison means -
+    profiles/spike_probe_r02.txt)."""
     for _ in range(12):
         fn()
-    for _ in range(3):
-        fn()
-    torch.cuda.synchronize(); t0 = time.perf_counter()
+    ts = []
     for _ in range(iters):
+        torch.cuda.synchronize(); t0 = time.perf_counter()
         fn()
-    torch.cuda.synchronize()
-    return (time.perf_counter() - t0) / iters * 1e6
+        torch.cuda.synchronize()
+        ts.append((time.perf_counter() - t0) * 1e6)
+    return sorted(ts)[iters // 2]
 
 print(f"{'shape':10s} {'M':>5} {'N':>5} {'K':>5} | {'ours':>8} {'blaslt':>8} us | ours/lib  TF(lib)")
 for name, m, n, k in shapes:
