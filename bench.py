@@ -155,9 +155,17 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     if rank == 0:
         images = args.batch_size * args.steps
         value = images / elapsed
+        # BASELINE.json metric name for the flagship config; non-default
+        # models label their own items/sec honestly
+        if model_name == "google/vit-base-patch16-224":
+            metric = ("images/sec (whole node) ViT-Base/224 microbatch=8 "
+                      "at 1/2/4/8-stage pipeline")
+        else:
+            unit_kind = "items" if config.model_type == "bert" else "images"
+            metric = (f"{unit_kind}/sec (whole node) {model_name} "
+                      f"microbatch={args.ubatch_size} pipeline")
         out = {
-            "metric": "images/sec (whole node) ViT-Base/224 microbatch=8 "
-                      "at 1/2/4/8-stage pipeline",
+            "metric": metric,
             "value": round(value, 3),
             "unit": "images/s",
             "n_gpus": world_size,
