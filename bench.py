@@ -167,7 +167,7 @@ def run_rank(args, rank: int, world_size: int) -> dict:
         out = {
             "metric": metric,
             "value": round(value, 3),
-            "unit": "images/s",
+            "unit": "items/s" if config.model_type == "bert" else "images/s",
             "n_gpus": world_size,
             "steps": args.steps,
             "warmup": args.warmup,
