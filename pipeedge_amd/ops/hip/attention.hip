@@ -326,12 +326,16 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
 // the dh==DHP guards away (the ViT/BERT dh=64 hot case). KROW pads the K
 // LDS row to 128 elements for DHP > 64 so the row stride stays 256 B and
 // the 16-chunk XOR swizzle keeps the A-fragment reads conflict-free.
-template <int KVT, int DHP = 64, bool EXACT = true, bool MASK = false>
-__global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
+// NW: waves per block (QB = NW*32 q rows). NW=2 halves the block so the
+// ragged S=197 tail wastes a 64-row block instead of half a 128-row one,
+// and 128-thread blocks pack the CUs finer.
+template <int KVT, int DHP = 64, bool EXACT = true, bool MASK = false,
+          int NW = 4>
+__global__ __launch_bounds__(64 * NW) void qkv_attention_v2_kernel(
         const __bf16* __restrict__ qkv, __bf16* __restrict__ ctx,
         const float* __restrict__ mask, int B, int S, int H, int dh,
         float scale) {
-    constexpr int QB = 128;                  // q rows per block (4 x 32)
+    constexpr int QB = NW * 32;              // q rows per block
     constexpr int KROW = (DHP <= 64) ? 64 : 128;   // padded K row (elems)
     constexpr int SWZ = (DHP <= 64) ? 7 : 15;      // 16-B chunk XOR mask
     // DHP need only be a 16-multiple: scores/Q/K/staging run at the NATIVE
@@ -377,15 +381,16 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     // bijective piece index (DHP/8 need not divide 256; the piece total
     // need not be a 256-multiple either - DHP=80 at KVT=64 has 640 pieces,
     // the top half of the last round is idle) ----
+    constexpr int NTHR = 64 * NW;
     constexpr int TOTAL_PIECES = KVT * DHP / 8;
-    constexpr int PIECES = (TOTAL_PIECES + 255) / 256;
-    constexpr bool RAGGED = (TOTAL_PIECES % 256) != 0;
+    constexpr int PIECES = (TOTAL_PIECES + NTHR - 1) / NTHR;
+    constexpr bool RAGGED = (TOTAL_PIECES % NTHR) != 0;
     int lr[PIECES], lc[PIECES];
     const __bf16* k_pc[PIECES];
     const __bf16* v_pc[PIECES];
     #pragma unroll
     for (int i = 0; i < PIECES; ++i) {
-        const int piece = tid + i * 256;
+        const int piece = tid + i * NTHR;
         const int p = RAGGED ? min(piece, TOTAL_PIECES - 1) : piece;
         lr[i] = p / (DHP / 8);
         lc[i] = (p % (DHP / 8)) * 8;
@@ -397,7 +402,7 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
         const int s0_ = t * KVT;
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            if (RAGGED && tid + i * 256 >= TOTAL_PIECES) continue;
+            if (RAGGED && tid + i * NTHR >= TOTAL_PIECES) continue;
             const int gs = s0_ + lr[i];
             k_reg[i] = bf16x8{};
             v_reg[i] = bf16x8{};
@@ -412,7 +417,7 @@ __global__ __launch_bounds__(256) void qkv_attention_v2_kernel(
     auto write_tile = [&]() {
         #pragma unroll
         for (int i = 0; i < PIECES; ++i) {
-            if (RAGGED && tid + i * 256 >= TOTAL_PIECES) continue;
+            if (RAGGED && tid + i * NTHR >= TOTAL_PIECES) continue;
             const int r = lr[i];
             // K: swizzle the 16-B column chunk by row&SWZ (reads match)
             const int kchunk = (lc[i] / 8) ^ (r & SWZ);
@@ -811,20 +816,23 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         const char* e = std::getenv("PIPEEDGE_ATTN_V2");
         return e ? std::atoi(e) : 1;
     }();
-// launch helpers: pick the MASK template arm from mask_ptr
-#define PE_LAUNCH_V2(KVT_, DHP_, EXACT_)                                     \
+// launch helpers: pick the MASK template arm from mask_ptr. NW_ = waves
+// per block (q rows = 32*NW_).
+#define PE_LAUNCH_V2_NW(KVT_, DHP_, EXACT_, NW_)                             \
     do {                                                                     \
+        dim3 gridv2(B * H, (S + 32 * NW_ - 1) / (32 * NW_));                 \
         if (mask_ptr)                                                        \
             hipLaunchKernelGGL(                                              \
-                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, true>), grid2,  \
-                dim3(256), 0, pe_current_stream(), in, out, mask_ptr, B, S,  \
-                H, dh, (float)scale);                                        \
+                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, true, NW_>),    \
+                gridv2, dim3(64 * NW_), 0, pe_current_stream(), in, out,     \
+                mask_ptr, B, S, H, dh, (float)scale);                        \
         else                                                                 \
             hipLaunchKernelGGL(                                              \
-                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, false>), grid2, \
-                dim3(256), 0, pe_current_stream(), in, out, nullptr, B, S,   \
-                H, dh, (float)scale);                                        \
+                (qkv_attention_v2_kernel<KVT_, DHP_, EXACT_, false, NW_>),   \
+                gridv2, dim3(64 * NW_), 0, pe_current_stream(), in, out,     \
+                nullptr, B, S, H, dh, (float)scale);                         \
     } while (0)
+#define PE_LAUNCH_V2(KVT_, DHP_, EXACT_) PE_LAUNCH_V2_NW(KVT_, DHP_, EXACT_, 4)
 #define PE_LAUNCH_V1(DHP_, KVT_, EXACT_)                                     \
     do {                                                                     \
         if (mask_ptr)                                                        \
@@ -840,7 +848,12 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     } while (0)
 
     if (dh == 64 && v2) {
-        dim3 grid2(B * H, (S + 127) / 128);
+        // PIPEEDGE_ATTN_NW=2 launches 128-thread 64-q-row blocks (finer
+        // packing, less ragged-tail waste at S=197) - A/B knob
+        static const int nw = []() {
+            const char* e = std::getenv("PIPEEDGE_ATTN_NW");
+            return e ? std::atoi(e) : 4;
+        }();
         // KV tile length: 128 halves the barrier count and wins SOLO at
         // ViT-class sequences (9.1 vs 9.7us), but 64's 16 KiB LDS doubles
         // co-residency (8 vs 5 blocks/CU) and wins the 2-stream e2e bench
@@ -850,6 +863,8 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         const bool kvt128 = v2 == 2;
         if (kvt128)
             PE_LAUNCH_V2(128, 64, true);
+        else if (nw == 2)
+            PE_LAUNCH_V2_NW(64, 64, true, 2);
         else
             PE_LAUNCH_V2(64, 64, true);
         HIP_CHECK_LAST();
@@ -859,14 +874,12 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
         // ViT-Huge: NATIVE DHP=80 scores/Q/K/staging (round 1 zero-padded
         // to 96: 20% wasted QK^T MFMAs + staging); KVT=64 keeps the 28 KiB
         // LDS footprint (5 blocks/CU)
-        dim3 grid2(B * H, (S + 127) / 128);
         PE_LAUNCH_V2(64, 80, false);
         HIP_CHECK_LAST();
         return ctx;
     }
     if (dh > 64 && dh <= 96 && dh % 8 == 0 && v2) {
         // other padded head dims: zero-padded to DHP=96
-        dim3 grid2(B * H, (S + 127) / 128);
         if (dh == 96)
             PE_LAUNCH_V2(64, 96, true);
         else
@@ -905,4 +918,5 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
     return ctx;
 }
 #undef PE_LAUNCH_V2
+#undef PE_LAUNCH_V2_NW
 #undef PE_LAUNCH_V1
