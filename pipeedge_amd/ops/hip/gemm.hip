@@ -734,6 +734,134 @@ __global__ __launch_bounds__(512) void gemm_bf16_glds8_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Phase-segregated 256x256 kernel (t16): the guide's 8-phase discipline
+// (cdna_hip_programming.md section 5 "256^2 8-phase template", measured
+// 1320-1340 TF random-data vs ~1100 for the 2-barrier structure). Each
+// K-tile is computed as 4 quadrant-phases; within a phase every wave
+// posts its quadrant's 12 ds_read_b128 + 2 glds, hits a raw barrier +
+// lgkmcnt(0), then issues 16 MFMAs under setprio(1) - reads and MFMAs are
+// SEGREGATED so the matrix pipe runs the MFMA segment back-to-back for
+// both co-resident waves instead of stalling on interleaved dependencies.
+// Tile t+1 stages into the buffer not being read (2 glds/wave/phase); the
+// tile boundary pays one vmcnt(0) for the phase-3-issued glds (a
+// conservative deviation from the template's vmcnt(6), which needs a
+// 2-tile staging lead the 2-buffer LDS cannot hold).
+// ---------------------------------------------------------------------------
+// PHV 0: 2 glds/phase, 2 barriers/phase, boundary vmcnt(0).
+// PHV 1: glds front-loaded into phases 0-1 (boundary waits 2-phase-old
+//        loads instead of just-issued ones).
+// PHV 2: PHV1 + phase-end barrier dropped (reads of the next quadrant
+//        overlap the partner wave's MFMA tail).
+template <int PHV = 0>
+__global__ __launch_bounds__(512) void gemm_bf16_phase8_kernel(
+        const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+        __bf16* __restrict__ C, int M, int N, int K, EpilogueArgs ep) {
+    constexpr int MI = 8, NI = 4;
+    constexpr int TBM = 256, TBN = 256, KB = 64;
+    __shared__ __bf16 smem[2][TBM + TBN][KB];
+
+    int bx = blockIdx.x, by = blockIdx.y;
+    {
+        const int nb = gridDim.x * gridDim.y;
+        const int id = xcd_swizzle(by * gridDim.x + bx, nb);
+        bx = id % gridDim.x;
+        by = id / gridDim.x;
+    }
+    const int bm = bx * TBM;
+    const int bn = by * TBN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wm = (wid >> 2) * (16 * MI);
+    const int wn = (wid & 3) * (16 * NI);
+
+    // staging: 8 glds per wave per tile, 2 per phase
+    const int srow = lane >> 3;
+    const int cdst = lane & 7;
+    const __bf16* a_pc[4];
+    const __bf16* b_pc[4];
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        const int p = wid * 4 + i;
+        const int trow = p * 8 + srow;
+        const int csrc = cdst ^ (trow & 7);
+        a_pc[i] = A + (long)min(bm + trow, M - 1) * K + csrc * 8;
+        b_pc[i] = B + (long)min(bn + trow, N - 1) * K + csrc * 8;
+    }
+    // glds pair g (0..3) of buffer `buf`: A piece g for g<... interleave
+    // A/B so each phase stages one A and one B piece
+    auto stage2 = [&](int buf, int g) {
+        const int p = wid * 4 + g;
+        glds_b128(a_pc[g], &smem[buf][p * 8][0]);
+        a_pc[g] += KB;
+        glds_b128(b_pc[g], &smem[buf][TBM + p * 8][0]);
+        b_pc[g] += KB;
+    };
+
+    f32x4 acc[MI][NI] = {};
+    const int n_tiles = K / KB;
+    // prologue: stage tile 0 fully, drain
+    #pragma unroll
+    for (int g = 0; g < 4; ++g) stage2(0, g);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int t = 0; t < n_tiles; ++t) {
+        const int buf = t & 1;
+        const bool more = t + 1 < n_tiles;
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            // quadrant q = (mh, nh): M-frags mh*4.., N-frags nh*2..
+            const int mh = q >> 1, nh = q & 1;
+            bf16x8 af[4][2], bfr[2][2];
+            #pragma unroll
+            for (int kk = 0; kk < 2; ++kk) {
+                const int kgrp = kk * 32 + (lane >> 4) * 8;
+                #pragma unroll
+                for (int mi = 0; mi < 4; ++mi)
+                    af[mi][kk] = lds_read_swz<KB>(
+                        &smem[buf][0][0],
+                        wm + (mh * 4 + mi) * 16 + (lane & 15), kgrp);
+                #pragma unroll
+                for (int ni = 0; ni < 2; ++ni)
+                    bfr[ni][kk] = lds_read_swz<KB>(
+                        &smem[buf][TBM][0],
+                        wn + (nh * 2 + ni) * 16 + (lane & 15), kgrp);
+            }
+            if (more) {
+                if (PHV == 0) {
+                    stage2(buf ^ 1, q);
+                } else if (q < 2) {
+                    stage2(buf ^ 1, 2 * q);
+                    stage2(buf ^ 1, 2 * q + 1);
+                }
+            }
+            __builtin_amdgcn_s_barrier();
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_setprio(1);
+            #pragma unroll
+            for (int kk = 0; kk < 2; ++kk)
+                #pragma unroll
+                for (int mi = 0; mi < 4; ++mi)
+                    #pragma unroll
+                    for (int ni = 0; ni < 2; ++ni)
+                        acc[mh * 4 + mi][nh * 2 + ni] = mfma_16x16x32_bf16(
+                            af[mi][kk], bfr[ni][kk],
+                            acc[mh * 4 + mi][nh * 2 + ni]);
+            __builtin_amdgcn_s_setprio(0);
+            if (PHV < 2)
+                __builtin_amdgcn_s_barrier();
+        }
+        // tile boundary: next phase reads buf^1; its glds were issued
+        // during this tile's phases (latest in phase 3)
+        if (more)
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+    gemm_epilogue<MI, NI, 0>(acc, ep, bm, bn, wm, wn, lane, 0, nullptr);
+}
+
+// ---------------------------------------------------------------------------
 // generic path: any K (padded LDS, plain vectorized staging)
 // ---------------------------------------------------------------------------
 template <int EPM = 0>
@@ -984,6 +1112,22 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                                    dim3(512), 0, pe_current_stream(), ap, bp,
                                    cp, M, N, K, ep);
             }
+        } else if (t == 16 && epm == 0) {
+            dim3 grid((M + 255) / 256, (N + 255) / 256);
+            const char* pv = std::getenv("PIPEEDGE_P8V");
+            const int phv = pv ? std::atoi(pv) : 1;
+            if (phv == 2)
+                hipLaunchKernelGGL((gemm_bf16_phase8_kernel<2>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
+            else if (phv == 0)
+                hipLaunchKernelGGL((gemm_bf16_phase8_kernel<0>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
+            else
+                hipLaunchKernelGGL((gemm_bf16_phase8_kernel<1>), grid,
+                                   dim3(512), 0, pe_current_stream(), ap, bp,
+                                   cp, M, N, K, ep);
         } else if (t == 15 && epm == 0) {
             // 256x192 2-buffer (112 KiB LDS): ViT-H fc1 grid = 9x27 = 243
             // blocks ~ one full chip wave (256x256 gives only 180)
@@ -992,7 +1136,7 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                                dim3(512), 0, pe_current_stream(), ap, bp,
                                cp, M, N, K, ep);
         } else if (t == 7 || t == 12 || t == 13 || t == 14 || t == 15 ||
-                   t == 20) {
+                   t == 16 || t == 20) {
             dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
             PE_LAUNCH_GLDS3(2, 2, 3, grid);
         } else if (t == 10) {
