@@ -535,15 +535,25 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
     with DistP2pContext(world_size, rank, handle_cmd) as dist_ctx:
         # distribute the schedule (reference runtime.py:431-448)
         if rank == 0:
-            stage_layers, stage_quant, stage_ranks = get_pipeline_sched(
-                world_size, hosts, partition, quant_list, rank_order,
-                model_name, ubatch_size, dtype_str, sched_models_file,
-                sched_dev_types_file, sched_dev_file)
-            # seq_len: data-dependent for text models; broadcast with schedule
-            seq_len = 0
-            if model_cfg.get_model_config(model_name).model_type == "bert":
-                dataset_probe = load_dataset(dataset_cfg, model_name, 1, 1)
-                seq_len = dataset_probe[0][0].shape[-1]
+            try:
+                stage_layers, stage_quant, stage_ranks = get_pipeline_sched(
+                    world_size, hosts, partition, quant_list, rank_order,
+                    model_name, ubatch_size, dtype_str, sched_models_file,
+                    sched_dev_types_file, sched_dev_file)
+                # seq_len: data-dependent for text models; broadcast with
+                # the schedule
+                seq_len = 0
+                if model_cfg.get_model_config(model_name).model_type == "bert":
+                    dataset_probe = load_dataset(dataset_cfg, model_name, 1, 1)
+                    seq_len = dataset_probe[0][0].shape[-1]
+            except Exception:
+                # a scheduling failure must not strand the other ranks in
+                # their schedule wait forever (observed: a bad YAML hung a
+                # whole 8-rank job until an external timeout killed it)
+                logger.exception("Scheduling failed; broadcasting stop")
+                dist_ctx.cmd_broadcast(CMD_STOP)
+                stop_event.set()
+                raise
             logger.info("Scheduling: data rank: %s", data_rank)
             dist_ctx.cmd_broadcast(CMD_SCHED, (
                 torch.tensor(stage_layers),
@@ -553,8 +563,16 @@ def run_pipeline_p2p(world_size: int, rank: int, model_name: str,
                 torch.tensor(seq_len)))
         else:
             logger.info("Waiting for schedule")
-            stage_layers, stage_quant, stage_ranks, data_rank, seq_len = \
-                sched_q.get()
+            while True:
+                try:
+                    sched = sched_q.get(timeout=0.25)
+                    break
+                except queue.Empty:
+                    if stop_event.is_set():
+                        raise RuntimeError(
+                            "stopped before a schedule arrived (rank 0 "
+                            "failed or the run was aborted)") from None
+            stage_layers, stage_quant, stage_ranks, data_rank, seq_len = sched
             stage_layers = [tuple(sl) for sl in stage_layers]
 
         try:

@@ -209,3 +209,31 @@ def test_runtime_bert_variable_length_npz(tmp_path):
     assert outs[0][0] == 0, outs[0][2]
     assert outs[1][0] == 0, outs[1][2]
     assert "throughput is" in outs[0][1]
+
+
+def test_runtime_sched_failure_does_not_hang(tmp_path):
+    """A rank-0 scheduling failure must abort ALL ranks promptly (observed:
+    a bad YAML left the other ranks waiting on the schedule forever)."""
+    import time
+    # device_types.yml has no profile for this model -> no viable schedule
+    bad_dev_types = {"mi355x": {"mem_MB": 1000, "bw_Mbps": 1000,
+                                "model_profiles": {}}}
+    model = {"layers": 48, "parameters_in": 1, "parameters_out": [1] * 48,
+             "mem_MB": [1.0] * 48}
+    with open(tmp_path / "models.yml", "w") as f:
+        yaml.safe_dump({MODEL: model}, f)
+    with open(tmp_path / "device_types.yml", "w") as f:
+        yaml.safe_dump(bad_dev_types, f)
+    with open(tmp_path / "devices.yml", "w") as f:
+        yaml.safe_dump({"mi355x": ["0", "1"]}, f)
+    t0 = time.time()
+    outs = _run_two(29645, ["-b", "4", "-u", "2", "-m", MODEL,
+                            "-sm", str(tmp_path / "models.yml"),
+                            "-sdt", str(tmp_path / "device_types.yml"),
+                            "-sd", str(tmp_path / "devices.yml")],
+                    cwd=str(tmp_path))
+    elapsed = time.time() - t0
+    assert outs[0][0] != 0          # rank 0 reports the failure
+    assert elapsed < 120, f"ranks hung for {elapsed:.0f}s"
+    # rank 1 exited (either cleanly stopped or with the stop error)
+    assert outs[1][0] is not None
