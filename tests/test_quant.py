@@ -78,3 +78,19 @@ def test_quant_wire_words_formula():
     wire = model_spec.quant_wire_spec(specs, 3)
     per_elem = 197 * 768
     assert wire[0].shape == (1, math.ceil(per_elem / 10) * 4)
+
+
+def test_quant_roundtrip_additive_mask_tensor():
+    """BERT boundary payloads now include the additive padding mask
+    (0 / -1e9 values): the wire must carry its extremes exactly enough
+    that masked positions stay hugely negative and live positions ~0."""
+    import torch
+    from pipeedge_amd.quant import tensor_encode_outerdim, tensor_decode_outerdim
+    mask = torch.zeros(4, 128)
+    mask[:, 100:] = -1e9
+    for bit in (2, 4, 8):
+        wire = tensor_encode_outerdim(mask.clone(), bit)
+        out = tensor_decode_outerdim(wire)
+        assert out.shape == mask.shape
+        assert (out[:, 100:] < -1e8).all(), bit
+        assert (out[:, :100].abs() < 1e7).all(), bit

@@ -237,3 +237,27 @@ def test_runtime_sched_failure_does_not_hang(tmp_path):
     assert elapsed < 120, f"ranks hung for {elapsed:.0f}s"
     # rank 1 exited (either cleanly stopped or with the stop error)
     assert outs[1][0] is not None
+
+
+def test_runtime_bert_quantized_masked(tmp_path):
+    """Quantized wire x padding mask interplay: 2-stage BERT-CoLA with
+    -q 8,0 over a variable-length padded batch (the mask tensor rides the
+    quantized boundary)."""
+    import numpy as np
+    rng = np.random.default_rng(5)
+    n, smax = 8, 48
+    ids = np.zeros((n, smax), dtype=np.int64)
+    for i in range(n):
+        ln = int(rng.integers(8, smax + 1))
+        ids[i, :ln] = rng.integers(1000, 30000, size=ln)
+    np.savez(tmp_path / "cola.npz", ids=ids,
+             label=rng.integers(0, 2, size=n))
+    outs = _run_two(29647, ["-b", "8", "-u", "2",
+                            "-m", "textattack/bert-base-uncased-CoLA",
+                            "-pt", "1,24,25,48", "-q", "8,0",
+                            "--dataset-name", "npz",
+                            "--dataset-root", str(tmp_path / "cola.npz")],
+                    cwd=str(tmp_path))
+    assert outs[0][0] == 0, outs[0][2]
+    assert outs[1][0] == 0, outs[1][2]
+    assert "throughput is" in outs[0][1]
