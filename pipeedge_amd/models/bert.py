@@ -214,8 +214,19 @@ class BertModelShard(ModuleShard):
         elif isinstance(data, tuple):
             *inner, attn_mask = data
             data = inner[0] if len(inner) == 1 else tuple(inner)
+        kernel_mask = attn_mask
+        if attn_mask is not None and attn_mask.is_cuda:
+            # pre-pad ONCE per shard forward to the HIP kernel's f32
+            # [B, ceil128(S)] contract; every block reuses it (the per-call
+            # pad inside the op cost 3 extra launches per attention)
+            b, s = attn_mask.shape
+            sp = (s + 127) & ~127
+            km = torch.zeros(b, sp, dtype=torch.float32,
+                             device=attn_mask.device)
+            km[:, :s] = attn_mask.float()
+            kernel_mask = km
         for layer in self.layers:
-            data = layer(data, attn_mask)
+            data = layer(data, kernel_mask)
         if self.shard_config.is_last:
             # pooler: tanh(dense(CLS token))
             return ops.linear(data[:, 0], self.pooler_weight,

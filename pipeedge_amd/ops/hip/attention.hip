@@ -788,18 +788,27 @@ torch::Tensor qkv_attention_bf16(torch::Tensor qkv, int64_t num_heads,
 
     // Additive per-(batch, kpos) mask [B, S]: padded to a 128 multiple so
     // kernel tile loads are unguarded (pad positions add 0 and are then
-    // overridden by the ragged-tail -1e30 masking).
+    // overridden by the ragged-tail -1e30 masking). A caller may pass an
+    // ALREADY-PADDED f32 [B, ceil128(S)] mask (models pad once per shard
+    // forward and reuse it across their blocks - the per-call zeros+copy
+    // was 3 extra launches per attention).
     const float* mask_ptr = nullptr;
     torch::Tensor mask_padded;  // keeps the padded copy alive
     if (attn_mask.has_value()) {
         auto m = attn_mask.value();
         TORCH_CHECK(m.is_cuda(), "attention: mask must be a GPU tensor");
-        TORCH_CHECK(m.dim() == 2 && m.size(0) == B && m.size(1) == S,
-                    "attention: mask must be [B, S] (per-kpos additive)");
         const long sp = (S + 127) & ~127L;
-        mask_padded = torch::zeros({B, sp},
-                                   m.options().dtype(torch::kFloat32));
-        mask_padded.narrow(1, 0, S).copy_(m.to(torch::kFloat32));
+        if (m.dim() == 2 && m.size(0) == B && m.size(1) == sp &&
+            m.dtype() == torch::kFloat32 && m.is_contiguous()) {
+            mask_padded = m;  // pre-padded fast path
+        } else {
+            TORCH_CHECK(m.dim() == 2 && m.size(0) == B && m.size(1) == S,
+                        "attention: mask must be [B, S] (per-kpos additive)"
+                        " or pre-padded f32 [B, ceil128(S)]");
+            mask_padded = torch::zeros({B, sp},
+                                       m.options().dtype(torch::kFloat32));
+            mask_padded.narrow(1, 0, S).copy_(m.to(torch::kFloat32));
+        }
         mask_ptr = mask_padded.data_ptr<float>();
     }
     static const int kvt = []() {
