@@ -23,14 +23,12 @@ CPU tensors for CMD_SCHED / CMD_BATCH / CMD_STOP (reference runtime.py:36-44).
 import collections
 import logging
 import threading
-import time
 from typing import Callable, List, Optional, Sequence
 
 import torch
 import torch.distributed as dist
 
-from . import (CMD_BATCH, CMD_SCHED, CMD_STOP, DistCmdHandler, DistContext,
-               TensorSpec)
+from . import CMD_STOP, DistCmdHandler, DistContext, TensorSpec
 from ..models import ModuleShardData
 
 logger = logging.getLogger(__name__)

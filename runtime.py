@@ -19,7 +19,7 @@ import queue
 import sys
 import threading
 import time
-from typing import List, Optional, Tuple, Union
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch

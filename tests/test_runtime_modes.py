@@ -5,7 +5,6 @@ import os
 import subprocess
 import sys
 
-import pytest
 import yaml
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))

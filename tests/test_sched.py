@@ -1,7 +1,6 @@
 """Scheduler tests: YAML type contracts (mirrors reference
 test/sched/test_yaml_types.py), cost models, and golden-output tests of the
 optimal-throughput DP on hand-computable inputs."""
-import pytest
 
 from pipeedge_amd import sched
 from pipeedge_amd.sched import yaml_types
