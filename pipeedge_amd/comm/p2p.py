@@ -335,6 +335,28 @@ class DistP2pPipelineStage:
             return contextlib.nullcontext()
         return torch.cuda.stream(streams[i % len(streams)])
 
+    _offset_done = False
+
+    def _maybe_offset_stream(self, streams, i):
+        """One-time de-phasing of the later microbatch streams: without it
+        both streams start IDENTICAL per-microbatch kernel chains at the
+        same instant and stay phase-locked - both sit in the low-occupancy
+        LayerNorm/attention phases simultaneously. A one-time spin delay of
+        ~half a chain pairs one stream's GEMM phases against the other's
+        small-kernel phases. PIPEEDGE_STREAM_OFFSET_US (default 0 = off)."""
+        import os
+        if streams is None or self._offset_done or i == 0:
+            return
+        if i >= len(streams):
+            self._offset_done = True
+            return
+        us = int(os.environ.get("PIPEEDGE_STREAM_OFFSET_US", "0"))
+        if us > 0:
+            # ~2.0e3 device cycles per us on MI355X; exactness irrelevant
+            torch.cuda._sleep(int(us * 2000) * i)
+        if i == len(streams) - 1:
+            self._offset_done = True
+
     # -- stage-rank batch processing -----------------------------------------
     def _from_wire(self, tensors: List[torch.Tensor]) -> List[torch.Tensor]:
         if self.wire_device == self.device:
@@ -360,6 +382,7 @@ class DistP2pPipelineStage:
             recv.post(depth, n_ubatch)
         for i in range(n_ubatch):
             with self._stream_ctx(streams, i):
+                self._maybe_offset_stream(streams, i)
                 self._call_hook(self._recv_pre_hook)
                 wire_in = self._from_wire(recv.wait(i))
                 self._call_hook(self._recv_post_hook, wire_in)
@@ -387,6 +410,7 @@ class DistP2pPipelineStage:
             streams = self._ustreams()
             for i, ub in enumerate(ubatches):
                 with self._stream_ctx(streams, i):
+                    self._maybe_offset_stream(streams, i)
                     self.results_cb(self.work_cb(self._to_device(ub)))
             if streams is not None:
                 for s in streams:
