@@ -1021,8 +1021,11 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
             }();
             if (M >= 3072 && N >= 3072 && K >= 2048) t = 13;
             else if (M >= 2048) t = (N >= 3072) ? 15 : 5;
-            else if (M >= 1024 && N >= 4096 && K < 2048) t = 14;
-            else if (K >= 2048 || (N >= 3072 && M >= 1024)) t = 7;
+            // the wide-N branches below were tuned at M=1576 (ViT); at
+            // M=1024 (BERT u8 s128) t5 wins them (bertL-fc1: 28.9us vs
+            // t14 37.9 / t7 31.7 - profiles/bertL_probe.txt)
+            else if (M >= 1280 && N >= 4096 && K < 2048) t = 14;
+            else if (K >= 2048 || (N >= 3072 && M >= 1280)) t = 7;
             else if (midt != 0 && M >= 256) t = midt;
             else t = 5;
         }
