@@ -1020,6 +1020,12 @@ std::tuple<torch::Tensor, torch::Tensor> gemm_bf16(
                 return s ? std::atoi(s) : 0;
             }();
             if (M >= 3072 && N >= 3072 && K >= 2048) t = 13;
+            // t15 (256x192) for the wide-N M>=2048 class: solo probes say
+            // t7 is faster at shallow K (s384-fc1 44.8 vs 53.6us,
+            // profiles/m3072_probe.txt) but the s384 e2e bench REGRESSED
+            // 6.1k -> 5.7k items/s with t7 there - t15's single-block/CU
+            // footprint co-schedules better with the second microbatch
+            // stream. e2e wins over solo: keep t15.
             else if (M >= 2048) t = (N >= 3072) ? 15 : 5;
             // the wide-N branches below were tuned at M=1576 (ViT); at
             // M=1024 (BERT u8 s128) t5 wins them (bertL-fc1: 28.9us vs
