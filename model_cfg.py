@@ -207,8 +207,13 @@ def dist_p2p_pipeline_stage_factory(stage_ranks: List[int], data_rank: int,
         send_specs = edges[stage + 1]
         if stage == num_stages - 1 and rank_dst == data_rank:
             send_group = results_group  # back-edge sender side
-    return p2p.DistP2pPipelineStage(rank_src, rank_dst, recv_specs, send_specs,
-                                    work_cb, results_cb,
-                                    device=devices.DEVICE,
-                                    recv_group=recv_group,
-                                    send_group=send_group)
+    ctx = p2p.DistP2pPipelineStage(rank_src, rank_dst, recv_specs, send_specs,
+                                   work_cb, results_cb,
+                                   device=devices.DEVICE,
+                                   recv_group=recv_group,
+                                   send_group=send_group)
+    # admission-control depth for the data rank (see DistP2pPipelineStage
+    # data_run_batch: the results back-edge makes unbounded admission a
+    # distributed-deadlock risk)
+    ctx.pipeline_depth = num_stages
+    return ctx

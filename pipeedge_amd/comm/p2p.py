@@ -422,7 +422,18 @@ class DistP2pPipelineStage:
         streams = self._ustreams()
         recv.post(2, n)
         collected = 0
+        # ADMISSION CONTROL - the results back-edge closes a CYCLE (data
+        # rank -> stages -> data rank). If more microbatches are admitted
+        # than the chain can buffer (~stages x (send window 8 + recv slots
+        # 2)), every rank ends up blocked on a full send window while the
+        # data rank itself is blocked sending instead of draining results:
+        # a distributed deadlock (observed at ~83 in-flight on an 8-stage
+        # continuous stream). Cap admissions well under capacity - steady
+        # state only needs ~2x pipeline depth in flight.
+        depth_cap = max(8, 4 * int(getattr(self, "pipeline_depth", 8)))
         for i, ub in enumerate(ubatches):
+            while i - collected >= depth_cap:
+                collected = self._consume_result(recv, collected, n)
             with self._stream_ctx(streams, i):
                 if self.work_cb is not None:
                     outputs = self.work_cb(self._to_device(ub))

@@ -143,3 +143,59 @@ def test_runtime_two_rank_cpu(extra):
     assert outs[0][0] == 0, outs[0][1][-2000:]
     assert outs[1][0] == 0, outs[1][1][-2000:]
     assert "throughput is" in outs[0][1]
+
+
+def _deep_stream_worker(rank, world_size, port, result_q):
+    """Stream FAR more microbatches than the chain can buffer: regression
+    for the results-back-edge deadlock (the cycle data rank -> stages ->
+    data rank saturated every send window at ~stages*(8+2) in-flight and
+    froze an 8-stage run; data-rank admission control bounds in-flight)."""
+    import logging
+    logging.disable(logging.WARNING)
+    _set_master(port)
+    from pipeedge_amd.comm.p2p import DistP2pContext, DistP2pPipelineStage
+
+    spec = [TensorSpec((1, 4), torch.float32)]
+    n_ub = 300   # >> capacity of a 4-stage chain (~40)
+    with DistP2pContext(world_size, rank, lambda c, t: None,
+                        backend="gloo"):
+        last = world_size - 1
+        if rank == 0:
+            results = []
+            stage = DistP2pPipelineStage(last, 1, spec, spec,
+                                         lambda x: x + 1.0, results.append,
+                                         device="cpu")
+            stage.pipeline_depth = world_size
+            ubatches = [torch.full((1, 4), float(i)) for i in range(n_ub)]
+            stage.data_run_batch(ubatches)
+            ok = len(results) == n_ub and all(
+                torch.allclose(results[i],
+                               torch.full((1, 4), i + float(world_size)))
+                for i in range(0, n_ub, 37))
+            result_q.put(("rank0", ok))
+        else:
+            nxt = 0 if rank == last else rank + 1
+            stage = DistP2pPipelineStage(rank - 1, nxt, spec, spec,
+                                         lambda x: x + 1.0, None,
+                                         device="cpu")
+            stage.stage_run_batch(n_ub)
+            result_q.put((f"rank{rank}", True))
+
+
+def test_deep_stream_no_deadlock():
+    ctx = mp.get_context("spawn")
+    result_q = ctx.Queue()
+    world = 4
+    procs = [ctx.Process(target=_deep_stream_worker,
+                         args=(r, world, 29617, result_q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        k, v = result_q.get(timeout=180)
+        results[k] = v
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert all(results.values()), results
