@@ -100,3 +100,24 @@ def test_bench_bert_four_stage_cpu():
     out = json.loads(lines[-1])
     assert out["n_gpus"] == 4
     assert out["config"]["seq_len"] == 128
+
+
+def test_bench_two_rank_deep_stream():
+    """Driver-shaped torchrun bench at a DEPTH that exceeds the 2-stage
+    chain's buffer capacity (~20 in-flight): regression for the results
+    back-edge deadlock the admission control fixes - the real SCALE run
+    streams steps x n_ubatch microbatches continuously."""
+    env = dict(os.environ, OMP_NUM_THREADS="2")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29671", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "40", "--warmup", "5",
+         "--batch-size", "8", "--ubatch-size", "2",
+         "--model", "facebook/deit-tiny-distilled-patch16-224"],
+        capture_output=True, timeout=600, env=env, cwd=REPO, check=False)
+    assert proc.returncode == 0, proc.stderr.decode()[-2000:]
+    line = [l for l in proc.stdout.decode().splitlines()
+            if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2 and out["steps"] == 40 and out["value"] > 0
