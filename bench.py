@@ -77,8 +77,12 @@ def run_rank(args, rank: int, world_size: int) -> dict:
     if distributed:
         dist.init_process_group(backend, rank=rank, world_size=world_size)
         # dedicated communicator for the results back-edge (see
-        # DistP2pPipelineStage group docs)
+        # DistP2pPipelineStage group docs); the barrier eagerly runs the
+        # COLLECTIVE NCCL comm init with every rank participating (only
+        # ranks 0 and N-1 ever issue ops on this group - lazy first-use
+        # init would hang)
         results_group = dist.new_group(backend=backend)
+        dist.barrier(group=results_group)
 
     model_name = args.model
     total_layers = model_cfg.get_model_layers(model_name)

@@ -126,6 +126,12 @@ class DistP2pContext(DistContext):
         # dedicated communicator for the results back-edge (see
         # DistP2pPipelineStage)
         self.results_group = dist.new_group(backend=self._backend)
+        # Eagerly initialize the subgroup communicator: NCCL comm init is
+        # COLLECTIVE over the group, but only the data rank and the last
+        # stage ever issue ops on this group - lazy first-use init would
+        # wait forever for the other ranks. A one-time barrier makes every
+        # rank participate in the init.
+        dist.barrier(group=self.results_group)
         self._cmd_thread = CommandThread(self._ctrl_group, self._cmd_cb,
                                          self.stop_event)
         self._cmd_thread.start()
