@@ -186,3 +186,18 @@ def test_bert_masked_stitched_equals_full(partition):
             cfg, _shard_cfg(a, b, 48), None)
         data = shard(data)
     assert torch.equal(data, y_full)
+
+
+def test_bert_mask_broadcast_equivalence():
+    """torch_ref [B,S] mask must equal an explicit [B,1,1,S] extended mask."""
+    from pipeedge_amd.ops import torch_ref
+    torch.manual_seed(2)
+    b, s, h, dh = 2, 32, 4, 16
+    qkv = torch.randn(b, s, 3 * h * dh)
+    mask = torch.zeros(b, s)
+    mask[0, 20:] = -1e9
+    mask[1, 10:] = -1e9
+    y2d = torch_ref.qkv_attention(qkv, h, attn_mask=mask)
+    y4d = torch_ref.qkv_attention(qkv, h,
+                                  attn_mask=mask.reshape(b, 1, 1, s))
+    assert torch.allclose(y2d, y4d)
